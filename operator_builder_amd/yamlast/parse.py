@@ -1,0 +1,314 @@
+"""Parse YAML into the comment-preserving AST.
+
+Strategy: PyYAML's composer builds the node graph (with source marks and
+scalar styles); a second pass over the raw text recovers comments from the
+gaps between scanner tokens; positional rules then attach each comment to
+a node the way gopkg.in/yaml.v3 does:
+
+  - a comment trailing content on a line -> ``line_comment`` of the last
+    scalar ending on that line before the comment;
+  - a block of full-line comments -> ``head_comment`` of the next node
+    (the innermost node starting at the next content position, i.e. the
+    key scalar of a mapping pair);
+  - a trailing block with no following node (or one separated from the
+    following node by a blank line) -> ``foot_comment`` of the node that
+    precedes it.
+"""
+
+from __future__ import annotations
+
+import yaml
+
+from .node import (
+    DOCUMENT,
+    MAPPING,
+    Node,
+    SCALAR,
+    SEQUENCE,
+    TAG_MAP,
+    TAG_SEQ,
+)
+
+
+class YAMLParseError(ValueError):
+    pass
+
+
+_TAG_ABBREV = "tag:yaml.org,2002:"
+
+
+def _abbrev_tag(tag: str) -> str:
+    if tag and tag.startswith(_TAG_ABBREV):
+        return "!!" + tag[len(_TAG_ABBREV) :]
+    return tag or ""
+
+
+def _convert(pynode, seen=None) -> Node:
+    if seen is None:
+        seen = {}
+    if id(pynode) in seen:
+        # aliased node: duplicate its converted form (comments stay with
+        # the anchor occurrence)
+        import copy
+
+        return copy.deepcopy(seen[id(pynode)])
+
+    start, end = pynode.start_mark, pynode.end_mark
+    common = dict(
+        line=start.line,
+        column=start.column,
+        end_line=end.line,
+        end_column=end.column,
+        index=start.index,
+        end_index=end.index,
+    )
+
+    if isinstance(pynode, yaml.ScalarNode):
+        node = Node(
+            kind=SCALAR,
+            tag=_abbrev_tag(pynode.tag),
+            value=pynode.value,
+            style=pynode.style if pynode.style else None,
+            **common,
+        )
+    elif isinstance(pynode, yaml.SequenceNode):
+        node = Node(
+            kind=SEQUENCE,
+            tag=TAG_SEQ,
+            flow=bool(pynode.flow_style),
+            **common,
+        )
+        seen[id(pynode)] = node
+        node.content = [_convert(child, seen) for child in pynode.value]
+    elif isinstance(pynode, yaml.MappingNode):
+        node = Node(
+            kind=MAPPING,
+            tag=TAG_MAP,
+            flow=bool(pynode.flow_style),
+            **common,
+        )
+        seen[id(pynode)] = node
+        for k, v in pynode.value:
+            node.content.append(_convert(k, seen))
+            node.content.append(_convert(v, seen))
+    else:
+        raise YAMLParseError(f"unsupported node type {type(pynode)!r}")
+
+    seen.setdefault(id(pynode), node)
+
+    return node
+
+
+def _scan_comments(src: str) -> list[dict]:
+    """Find every comment in the source with its absolute position.
+
+    Comments live in the gaps between scanner tokens, so '#' characters
+    inside scalars never produce false positives.
+    """
+    spans = []
+    try:
+        for token in yaml.scan(src):
+            s, e = token.start_mark.index, token.end_mark.index
+            if e > s:
+                spans.append((s, e))
+    except yaml.YAMLError as err:
+        raise YAMLParseError(f"error scanning yaml, {err}") from err
+
+    spans.sort()
+    merged: list[list[int]] = []
+    for s, e in spans:
+        if merged and s <= merged[-1][1]:
+            merged[-1][1] = max(merged[-1][1], e)
+        else:
+            merged.append([s, e])
+
+    # line-start offsets for position math
+    line_starts = [0]
+    for i, ch in enumerate(src):
+        if ch == "\n":
+            line_starts.append(i + 1)
+
+    def pos_of(idx: int) -> tuple[int, int]:
+        import bisect
+
+        line = bisect.bisect_right(line_starts, idx) - 1
+        return line, idx - line_starts[line]
+
+    comments = []
+    gaps = []
+    prev = 0
+    for s, e in merged:
+        if s > prev:
+            gaps.append((prev, s))
+        prev = max(prev, e)
+    if prev < len(src):
+        gaps.append((prev, len(src)))
+
+    for gs, ge in gaps:
+        i = gs
+        while i < ge:
+            if src[i] == "#":
+                eol = src.find("\n", i)
+                if eol == -1 or eol > ge:
+                    eol = ge
+                line, col = pos_of(i)
+                text = src[i:eol].rstrip()
+                ls = line_starts[line]
+                full_line = src[ls:i].strip() == ""
+                comments.append(
+                    dict(
+                        index=i,
+                        line=line,
+                        column=col,
+                        text=text,
+                        full_line=full_line,
+                    )
+                )
+                i = eol
+            else:
+                i += 1
+
+    comments.sort(key=lambda c: c["index"])
+
+    return comments
+
+
+def _blank_line_between(src_lines, a: int, b: int) -> bool:
+    for line_no in range(a + 1, b):
+        if 0 <= line_no < len(src_lines) and src_lines[line_no].strip() == "":
+            return True
+    return False
+
+
+def parse_documents(src: str) -> list[Node]:
+    """Parse a (possibly multi-document) YAML string into document Nodes
+    with comments attached."""
+    try:
+        pydocs = list(yaml.compose_all(src, Loader=yaml.SafeLoader))
+    except yaml.YAMLError as err:
+        raise YAMLParseError(f"error unmarshaling yaml, {err}") from err
+
+    docs = []
+    for pydoc in pydocs:
+        if pydoc is None:
+            continue
+        root = _convert(pydoc)
+        doc = Node(
+            kind=DOCUMENT,
+            line=root.line,
+            column=root.column,
+            end_line=root.end_line,
+            end_column=root.end_column,
+            index=root.index,
+            end_index=root.end_index,
+        )
+        doc.content = [root]
+        docs.append(doc)
+
+    comments = _scan_comments(src)
+    if comments:
+        _attach_comments(src, docs, comments)
+
+    return docs
+
+
+def _attach_comments(src: str, docs: list[Node], comments: list[dict]):
+    src_lines = src.split("\n")
+
+    # collect candidate nodes: all nodes of all documents, with depth so
+    # "innermost at a position" is resolvable
+    nodes: list[tuple[Node, int]] = []
+
+    def collect(node: Node, depth: int):
+        nodes.append((node, depth))
+        for child in node.content:
+            collect(child, depth + 1)
+
+    for doc in docs:
+        if doc.root is not None:
+            collect(doc.root, 0)
+
+    if not nodes:
+        return
+
+    scalars = [n for n, _ in nodes if n.kind == SCALAR]
+
+    # ---- line comments -------------------------------------------------
+    trailing = [c for c in comments if not c["full_line"]]
+    for c in trailing:
+        best = None
+        for n in scalars:
+            if n.end_line == c["line"] and n.end_index <= c["index"]:
+                if best is None or n.end_index > best.end_index:
+                    best = n
+        if best is None:
+            # e.g. a comment after a flow collection or on a key line
+            for n, _ in nodes:
+                if n.end_line == c["line"] and n.end_index <= c["index"]:
+                    if best is None or n.end_index > best.end_index:
+                        best = n
+        if best is not None:
+            if best.line_comment:
+                best.line_comment += "\n" + c["text"]
+            else:
+                best.line_comment = c["text"]
+
+    # ---- full-line comment blocks --------------------------------------
+    blocks: list[list[dict]] = []
+    for c in comments:
+        if not c["full_line"]:
+            continue
+        if blocks and c["line"] == blocks[-1][-1]["line"] + 1:
+            blocks[-1].append(c)
+        else:
+            blocks.append([c])
+
+    for block in blocks:
+        first, last = block[0], block[-1]
+        text = "\n".join(c["text"] for c in block)
+
+        # next node strictly after the block
+        nxt = None
+        nxt_depth = -1
+        for n, depth in nodes:
+            if n.index > last["index"]:
+                if (
+                    nxt is None
+                    or n.index < nxt.index
+                    or (n.index == nxt.index and depth > nxt_depth)
+                ):
+                    nxt = n
+                    nxt_depth = depth
+
+        # previous node ending before the block
+        prev = None
+        for n, _ in nodes:
+            if n.kind == SCALAR and n.end_index <= first["index"]:
+                if prev is None or n.end_index > prev.end_index:
+                    prev = n
+
+        attach_head = nxt is not None and not _blank_line_between(
+            src_lines, last["line"], nxt.line
+        )
+
+        if attach_head:
+            if nxt.kind != SCALAR:
+                # prefer the innermost scalar starting at the same index
+                for n, _ in nodes:
+                    if n.index == nxt.index and n.kind == SCALAR:
+                        nxt = n
+                        break
+            if nxt.head_comment:
+                nxt.head_comment += "\n" + text
+            else:
+                nxt.head_comment = text
+        elif prev is not None:
+            if prev.foot_comment:
+                prev.foot_comment += "\n" + text
+            else:
+                prev.foot_comment = text
+        elif nxt is not None:
+            if nxt.head_comment:
+                nxt.head_comment += "\n" + text
+            else:
+                nxt.head_comment = text
