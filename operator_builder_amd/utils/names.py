@@ -84,6 +84,15 @@ def regular_plural(kind: str) -> str:
     if word in _UNCOUNTABLE:
         return word
 
+    # words already ending in a plural-looking "s" (but not "ss"/"us"/"is",
+    # which are singular endings: ingress, status, analysis) pass through
+    # unchanged — role-rule resources arrive pre-pluralized
+    if (
+        word.endswith("s")
+        and not word.endswith(("ss", "us", "is"))
+        and len(word) > 1
+    ):
+        return word
     if word.endswith(("s", "x", "z", "ch", "sh")):
         return word + "es"
     if word.endswith("y") and len(word) > 1 and word[-2] not in "aeiou":
