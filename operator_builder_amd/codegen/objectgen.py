@@ -136,9 +136,9 @@ def generate(manifest_yaml: str, var_name: str) -> str:
 
     out: list[str] = [
         f"var {var_name} = &unstructured.Unstructured{{\n",
-        "\t\tObject: ",
+        "\tObject: ",
     ]
-    _emit(root, 2, out)
-    out.append(",\n\t}")
+    _emit(root, 1, out)
+    out.append(",\n}")
 
     return "".join(out)

@@ -1,0 +1,158 @@
+"""Template machinery: file emission + marker-based code insertion.
+
+Re-implements the slice of kubebuilder's ``machinery`` package the
+reference depends on (SURVEY.md §7 hard part #2):
+
+  - ``File``: a fully rendered file with an IfExistsAction
+    (OVERWRITE / SKIP / ERROR);
+  - ``Fragments``: code fragments spliced into an existing file at
+    scaffold markers (comment lines like ``//+kubebuilder:scaffold:imports``),
+    inserted *before* the marker line and deduplicated against the file's
+    current content so repeated ``create api`` runs are idempotent;
+  - ``Scaffold``: executes a batch of the above against a base directory.
+"""
+
+from __future__ import annotations
+
+import enum
+import os
+from dataclasses import dataclass, field
+
+
+class ScaffoldError(Exception):
+    pass
+
+
+class IfExists(enum.Enum):
+    OVERWRITE = "overwrite"
+    SKIP = "skip"
+    ERROR = "error"
+
+
+@dataclass(frozen=True)
+class Marker:
+    """A scaffold marker comment, e.g. ``//+kubebuilder:scaffold:imports``."""
+
+    comment: str  # "//" or "#"
+    value: str  # e.g. "kubebuilder:scaffold:imports"
+
+    def __str__(self) -> str:
+        return f"{self.comment}+{self.value}"
+
+
+@dataclass
+class File:
+    path: str
+    content: str
+    if_exists: IfExists = IfExists.OVERWRITE
+
+
+@dataclass
+class Fragments:
+    """Code fragments to insert at markers inside an existing file."""
+
+    path: str
+    fragments: dict[Marker, list[str]] = field(default_factory=dict)
+    # if the target file does not exist, optionally create it with this
+    # content first (mirrors kubebuilder updaters that are also creators)
+    missing_file_content: str | None = None
+
+
+class Scaffold:
+    def __init__(self, base_dir: str):
+        self.base_dir = base_dir
+
+    def _full(self, path: str) -> str:
+        return os.path.join(self.base_dir, path)
+
+    def execute(self, *items) -> None:
+        for item in items:
+            if isinstance(item, File):
+                self._write_file(item)
+            elif isinstance(item, Fragments):
+                self._insert_fragments(item)
+            else:
+                raise ScaffoldError(f"unknown scaffold item {item!r}")
+
+    def _write_file(self, item: File) -> None:
+        full = self._full(item.path)
+
+        if os.path.exists(full):
+            if item.if_exists == IfExists.SKIP:
+                return
+            if item.if_exists == IfExists.ERROR:
+                raise ScaffoldError(
+                    f"failed to create {item.path}: file already exists"
+                )
+
+        os.makedirs(os.path.dirname(full) or ".", exist_ok=True)
+        with open(full, "w", encoding="utf-8") as f:
+            f.write(item.content)
+
+    def _insert_fragments(self, item: Fragments) -> None:
+        full = self._full(item.path)
+
+        if not os.path.exists(full):
+            if item.missing_file_content is None:
+                raise ScaffoldError(
+                    f"unable to insert fragments: {item.path} does not exist"
+                )
+            os.makedirs(os.path.dirname(full) or ".", exist_ok=True)
+            with open(full, "w", encoding="utf-8") as f:
+                f.write(item.missing_file_content)
+
+        with open(full, encoding="utf-8") as f:
+            content = f.read()
+
+        content = insert_code_fragments(content, item.fragments)
+
+        with open(full, "w", encoding="utf-8") as f:
+            f.write(content)
+
+
+def insert_code_fragments(
+    content: str, fragments: dict[Marker, list[str]]
+) -> str:
+    """Insert each fragment immediately before its marker line, skipping
+    fragments already present anywhere in the file (kubebuilder's
+    dedupe-on-insert semantics)."""
+    lines = content.split("\n")
+
+    for marker, frags in fragments.items():
+        marker_text = str(marker)
+
+        to_insert = []
+        for frag in frags:
+            if frag.rstrip("\n") and frag.rstrip("\n") in content:
+                continue
+            if frag not in to_insert:
+                to_insert.append(frag)
+
+        if not to_insert:
+            continue
+
+        out_lines: list[str] = []
+        inserted = False
+        for line in lines:
+            if not inserted and line.strip() in (
+                marker_text,
+                marker_text.replace(marker.comment + "+", marker.comment + " +"),
+            ):
+                # fragments inherit the marker line's indentation
+                # (kubebuilder machinery behavior)
+                indent = line[: len(line) - len(line.lstrip())]
+                for frag in to_insert:
+                    for frag_line in frag.rstrip("\n").split("\n"):
+                        out_lines.append(
+                            indent + frag_line if frag_line else frag_line
+                        )
+                inserted = True
+            out_lines.append(line)
+
+        # a marker absent from the file is skipped silently (kubebuilder
+        # machinery behavior; e.g. the version subcommand updater targets
+        # a marker its template never renders)
+        lines = out_lines
+        content = "\n".join(lines)
+
+    return content
