@@ -1,0 +1,180 @@
+#!/usr/bin/env python3
+"""Benchmark: operator code-generation throughput.
+
+The reference (vmware-tanzu-labs/operator-builder) is a CPU-only CLI code
+generator with no GPU code path, no tensors, and no collectives
+(BASELINE.json north_star); its headline metric is
+correctness-shaped: generated-operator source for the bundled fixture
+workloads, plus codegen wall-clock.  This benchmark measures THAT metric:
+one "step" is a complete `init` + `create api` generation of both fixture
+operators (a standalone workload and a 3-workload collection) into a
+fresh directory — the reference's `make func-test` unit of work.
+
+Scaling: with N ranks each rank runs independent generation steps (weak
+scaling — the tool itself is single-process by design, matching the
+reference; N parallel ranks model N concurrent CI generation jobs).
+The reported value is whole-job aggregate codegen runs/second.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import shutil
+import sys
+import tempfile
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from operator_builder_amd.cli.main import main as ob_main  # noqa: E402
+
+FIXTURES = os.path.join(
+    os.path.dirname(os.path.abspath(__file__)), "tests", "fixtures"
+)
+
+
+class _Quiet:
+    """Silence the CLI's progress prints inside the timed region so the
+    benchmark emits exactly one JSON line."""
+
+    def __enter__(self):
+        self._stdout = sys.stdout
+        sys.stdout = open(os.devnull, "w")
+        return self
+
+    def __exit__(self, *exc):
+        sys.stdout.close()
+        sys.stdout = self._stdout
+
+
+def one_step(scratch: str) -> None:
+    """One benchmark step: generate both fixture operators."""
+    for fixture, repo in (
+        ("standalone", "github.com/acme/bookstore"),
+        ("collection", "github.com/acme/platform"),
+    ):
+        workdir = os.path.join(scratch, fixture)
+        if os.path.exists(workdir):
+            shutil.rmtree(workdir)
+        os.makedirs(workdir)
+        shutil.copytree(
+            os.path.join(FIXTURES, fixture),
+            os.path.join(workdir, ".workloadConfig"),
+        )
+
+        cwd = os.getcwd()
+        os.chdir(workdir)
+        try:
+            with _Quiet():
+                rc = ob_main(
+                    [
+                        "init",
+                        "--workload-config",
+                        ".workloadConfig/workload.yaml",
+                        "--repo",
+                        repo,
+                    ]
+                )
+                assert rc == 0, f"init failed for {fixture}"
+                rc = ob_main(["create", "api"])
+                assert rc == 0, f"create api failed for {fixture}"
+        finally:
+            os.chdir(cwd)
+
+
+def run(args) -> None:
+    import torch
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    distributed = world_size > 1
+
+    use_cuda = torch.cuda.is_available()
+
+    if distributed:
+        import torch.distributed as dist
+
+        backend = "nccl" if use_cuda else "gloo"
+        if use_cuda:
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+        dist.init_process_group(backend=backend)
+
+    def barrier_sync():
+        if distributed:
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    scratch = tempfile.mkdtemp(prefix=f"obbench-r{rank}-")
+
+    try:
+        for _ in range(args.warmup):
+            one_step(scratch)
+
+        barrier_sync()
+        t0 = time.perf_counter()
+
+        for _ in range(args.steps):
+            one_step(scratch)
+
+        barrier_sync()
+        elapsed = time.perf_counter() - t0
+    finally:
+        shutil.rmtree(scratch, ignore_errors=True)
+
+    # take the max elapsed over ranks
+    if distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if use_cuda:
+            t = t.cuda()
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    value = (args.steps * world_size) / elapsed
+    ms_per_step = (elapsed / args.steps) * 1000.0
+
+    if rank == 0:
+        print(
+            json.dumps(
+                {
+                    "metric": "codegen_runs_per_s",
+                    "value": value,
+                    "unit": "operator-generations/s",
+                    "n_gpus": world_size,
+                    "steps": args.steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": ms_per_step,
+                    "higher_is_better": True,
+                    "scaling": "weak",
+                    "vs_baseline": None,
+                    "dtype": "fp32",
+                    "data": (
+                        "synthetic fixture workloads (standalone + "
+                        "3-workload collection), bundled in-repo"
+                    ),
+                    "config": {
+                        "model": "operator-builder codegen (init + create api)",
+                        "global_batch": 2 * world_size,
+                        "seq_len": 0,
+                        "parallelism": f"dp{world_size}",
+                    },
+                }
+            )
+        )
+
+    if distributed:
+        dist.destroy_process_group()
+
+
+def parse_args():
+    parser = argparse.ArgumentParser(description=__doc__)
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=20)
+    parser.add_argument("--warmup", type=int, default=3)
+    return parser.parse_args()
+
+
+if __name__ == "__main__":
+    run(parse_args())
