@@ -1,5 +1,5 @@
 """YAML -> Go object source generation."""
 
-from .objectgen import generate, GenerateError
+from .objectgen import generate, generate_node, GenerateError
 
-__all__ = ["generate", "GenerateError"]
+__all__ = ["generate", "generate_node", "GenerateError"]

@@ -130,7 +130,13 @@ def generate(manifest_yaml: str, var_name: str) -> str:
         raise GenerateError(
             f"expected exactly one yaml document, got {len(docs)}"
         )
-    root = docs[0].root
+    return generate_node(docs[0], var_name)
+
+
+def generate_node(doc: Node, var_name: str) -> str:
+    """Generate Go source from an already-parsed document node (hot path:
+    avoids re-parsing text the pipeline already holds as an AST)."""
+    root = doc.root if doc.kind == "document" else doc
     if root is None or root.kind != MAPPING:
         raise GenerateError("manifest root must be a mapping")
 

@@ -39,7 +39,13 @@ class Inspector:
         self, data: str, *transforms: YAMLTransformer
     ) -> tuple[list[Node], list[YAMLResult]]:
         docs = parse_documents(data)
+        return docs, self.inspect_parsed(docs, *transforms)
 
+    def inspect_parsed(
+        self, docs: list[Node], *transforms: YAMLTransformer
+    ) -> list[YAMLResult]:
+        """Inspect already-parsed document nodes (hot path: avoids
+        re-parsing text the pipeline already holds as an AST)."""
         results: list[YAMLResult] = []
         for doc in docs:
             results.extend(self._inspect_nodes(doc.content))
@@ -51,7 +57,7 @@ class Inspector:
         for transform in transforms:
             transform(*results)
 
-        return docs, results
+        return results
 
     def _inspect_nodes(self, nodes: list[Node]) -> list[YAMLResult]:
         results: list[YAMLResult] = []

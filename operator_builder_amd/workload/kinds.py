@@ -16,9 +16,25 @@ from __future__ import annotations
 from dataclasses import dataclass, field
 from typing import Any, Optional
 
-from ..codegen import generate
+from ..codegen import generate_node
 from ..utils import to_package_name, regular_plural
-from ..yamlast import emit_document, parse_documents, to_plain
+from ..yamlast import emit_document, to_plain
+from ..yamlast.node import SCALAR, TAG_VAR
+
+
+def _rewrite_collection_refs(doc) -> None:
+    """Convert collection-marker substitutions to parent substitutions on
+    a collection's own resources (a collection marker on a collection is
+    a field marker to itself)."""
+    for node in doc.walk():
+        if node.kind != SCALAR:
+            continue
+        if node.tag == TAG_VAR and node.value.startswith("collection."):
+            node.value = "parent." + node.value[len("collection.") :]
+        if "!!start collection." in node.value:
+            node.value = node.value.replace(
+                "!!start collection.", "!!start parent."
+            )
 from . import rbac
 from .api_fields import APIFields
 from .companion import CLI
@@ -214,20 +230,12 @@ class WorkloadSpec:
         unique_names: set[str] = set()
 
         for manifest_file in self.manifests:
-            self.process_markers(manifest_file, *marker_types)
+            docs = self.process_markers(manifest_file, *marker_types)
 
             child_resources: list[ChildResource] = []
 
-            for doc_text in manifest_file.extract_manifests():
-                try:
-                    docs = parse_documents(doc_text)
-                    obj = to_plain(docs[0]) if docs else None
-                except Exception as err:
-                    raise ProcessManifestError(
-                        f"{err}; error processing manifest file - unable to "
-                        f"decode object in manifest file "
-                        f"{manifest_file.filename}"
-                    ) from err
+            for doc in docs:
+                obj = to_plain(doc)
 
                 if not isinstance(obj, dict) or not obj.get("kind"):
                     raise ProcessManifestError(
@@ -247,8 +255,9 @@ class WorkloadSpec:
                     )
                 unique_names.add(child.unique_name)
 
-                child.source_code = generate(doc_text, "resourceObj")
-                child.static_content = doc_text
+                child.doc = doc
+                child.source_code = generate_node(doc, "resourceObj")
+                child.static_content = emit_document(doc)
 
                 child_resources.append(child)
 
@@ -258,7 +267,7 @@ class WorkloadSpec:
 
     def process_markers(
         self, manifest_file: Manifest, *marker_types: MarkerType
-    ) -> None:
+    ) -> list:
         try:
             nodes, marker_results = inspect_for_yaml(
                 manifest_file.content, *marker_types
@@ -269,21 +278,23 @@ class WorkloadSpec:
                 f"[{manifest_file.filename}]"
             ) from err
 
-        content = "".join("---\n" + emit_document(doc) for doc in nodes)
-        manifest_file.content = content
-
         self.process_marker_results(marker_results)
 
         # collection markers on collection resources are rewritten to field
         # markers so the generated code compiles (reference
-        # workload.go:313-328)
+        # workload.go:313-328) — applied on the AST, so downstream codegen
+        # and the re-emitted text both see the rewrite
         if MarkerType.FIELD in marker_types and (
             MarkerType.COLLECTION in marker_types
         ):
-            content = manifest_file.content
-            content = content.replace("!!var collection", "!!var parent")
-            content = content.replace("!!start collection", "!!start parent")
-            manifest_file.content = content
+            for doc in nodes:
+                _rewrite_collection_refs(doc)
+
+        manifest_file.content = "".join(
+            "---\n" + emit_document(doc) for doc in nodes
+        )
+
+        return nodes
 
     def process_marker_results(self, marker_results) -> None:
         for result in marker_results:

@@ -45,8 +45,9 @@ class ChildResource:
     source_code: str = ""
     include_code: str = ""
     rbac: rbac.Rules = field(default_factory=rbac.Rules)
-    # set during func-name dedup (manifest.go:122-160)
-    use_strconv: bool = False
+    # the parsed document node this resource came from (kept so resource
+    # markers can be re-discovered without re-parsing static_content)
+    doc: Optional[object] = None
 
     def __str__(self) -> str:
         return (
@@ -82,9 +83,16 @@ class ChildResource:
     def process_resource_markers(
         self, marker_collection: MarkerCollection
     ) -> None:
-        _, marker_results = inspect_for_yaml(
-            self.static_content, MarkerType.RESOURCE
-        )
+        if self.doc is not None:
+            from .markers import inspect_parsed_for_markers
+
+            marker_results = inspect_parsed_for_markers(
+                [self.doc], MarkerType.RESOURCE
+            )
+        else:
+            _, marker_results = inspect_for_yaml(
+                self.static_content, MarkerType.RESOURCE
+            )
 
         if not marker_results:
             return

@@ -361,6 +361,15 @@ def inspect_for_yaml(
     return inspector.inspect_yaml(yaml_content, transform_yaml)
 
 
+def inspect_parsed_for_markers(
+    docs: list[Node], *marker_types: MarkerType
+) -> list[YAMLResult]:
+    """Inspect already-parsed documents (used to re-discover resource
+    markers on child resources without re-parsing their text)."""
+    inspector = initialize_marker_inspector(*marker_types)
+    return inspector.inspect_parsed(docs, transform_yaml)
+
+
 # ---- transform ---------------------------------------------------------
 
 # markers reserved for internal purposes (markers.go:155-175)

@@ -1,9 +1,9 @@
 """Parse YAML into the comment-preserving AST.
 
 Strategy: PyYAML's composer builds the node graph (with source marks and
-scalar styles); a second pass over the raw text recovers comments from the
-gaps between scanner tokens; positional rules then attach each comment to
-a node the way gopkg.in/yaml.v3 does:
+scalar styles) through a Scanner subclass that records every comment with
+its position during the same single pass; positional rules then attach
+each comment to a node the way gopkg.in/yaml.v3 does:
 
   - a comment trailing content on a line -> ``line_comment`` of the last
     scalar ending on that line before the comment;
@@ -32,6 +32,47 @@ from .node import (
 
 class YAMLParseError(ValueError):
     pass
+
+
+class _CommentLoader(yaml.SafeLoader):
+    """SafeLoader that records every comment while scanning, so parsing
+    and comment discovery happen in ONE pass over the input (comments are
+    only ever consumed inside ``scan_to_next_token``)."""
+
+    def __init__(self, stream):
+        super().__init__(stream)
+        self.collected_comments: list[dict] = []
+
+    def scan_to_next_token(self):
+        # mirrors PyYAML Scanner.scan_to_next_token, adding comment capture
+        if self.index == 0 and self.peek() == "\ufeff":
+            self.forward()
+        found = False
+        while not found:
+            while self.peek() == " ":
+                self.forward()
+            if self.peek() == "#":
+                start_index = self.index
+                start_line = self.line
+                start_column = self.column
+                chars = []
+                while self.peek() not in "\0\r\n\x85\u2028\u2029":
+                    chars.append(self.peek())
+                    self.forward()
+                self.collected_comments.append(
+                    dict(
+                        index=start_index,
+                        line=start_line,
+                        column=start_column,
+                        text="".join(chars).rstrip(),
+                        full_line=False,  # fixed up by the caller
+                    )
+                )
+            if self.scan_line_break():
+                if not self.flow_level:
+                    self.allow_simple_key = True
+            else:
+                found = True
 
 
 _TAG_ABBREV = "tag:yaml.org,2002:"
@@ -99,78 +140,17 @@ def _convert(pynode, seen=None) -> Node:
     return node
 
 
-def _scan_comments(src: str) -> list[dict]:
-    """Find every comment in the source with its absolute position.
-
-    Comments live in the gaps between scanner tokens, so '#' characters
-    inside scalars never produce false positives.
-    """
-    spans = []
-    try:
-        for token in yaml.scan(src):
-            s, e = token.start_mark.index, token.end_mark.index
-            if e > s:
-                spans.append((s, e))
-    except yaml.YAMLError as err:
-        raise YAMLParseError(f"error scanning yaml, {err}") from err
-
-    spans.sort()
-    merged: list[list[int]] = []
-    for s, e in spans:
-        if merged and s <= merged[-1][1]:
-            merged[-1][1] = max(merged[-1][1], e)
-        else:
-            merged.append([s, e])
-
-    # line-start offsets for position math
+def _fixup_full_line(src: str, comments: list[dict]) -> None:
+    """Mark comments that have nothing but whitespace before them on
+    their line (head/foot comments vs trailing line comments)."""
     line_starts = [0]
     for i, ch in enumerate(src):
         if ch == "\n":
             line_starts.append(i + 1)
 
-    def pos_of(idx: int) -> tuple[int, int]:
-        import bisect
-
-        line = bisect.bisect_right(line_starts, idx) - 1
-        return line, idx - line_starts[line]
-
-    comments = []
-    gaps = []
-    prev = 0
-    for s, e in merged:
-        if s > prev:
-            gaps.append((prev, s))
-        prev = max(prev, e)
-    if prev < len(src):
-        gaps.append((prev, len(src)))
-
-    for gs, ge in gaps:
-        i = gs
-        while i < ge:
-            if src[i] == "#":
-                eol = src.find("\n", i)
-                if eol == -1 or eol > ge:
-                    eol = ge
-                line, col = pos_of(i)
-                text = src[i:eol].rstrip()
-                ls = line_starts[line]
-                full_line = src[ls:i].strip() == ""
-                comments.append(
-                    dict(
-                        index=i,
-                        line=line,
-                        column=col,
-                        text=text,
-                        full_line=full_line,
-                    )
-                )
-                i = eol
-            else:
-                i += 1
-
-    comments.sort(key=lambda c: c["index"])
-
-    return comments
+    for c in comments:
+        ls = line_starts[c["line"]] if c["line"] < len(line_starts) else 0
+        c["full_line"] = src[ls : c["index"]].strip() == ""
 
 
 def _blank_line_between(src_lines, a: int, b: int) -> bool:
@@ -183,10 +163,16 @@ def _blank_line_between(src_lines, a: int, b: int) -> bool:
 def parse_documents(src: str) -> list[Node]:
     """Parse a (possibly multi-document) YAML string into document Nodes
     with comments attached."""
+    loader = _CommentLoader(src)
     try:
-        pydocs = list(yaml.compose_all(src, Loader=yaml.SafeLoader))
+        pydocs = []
+        while loader.check_node():
+            pydocs.append(loader.get_node())
     except yaml.YAMLError as err:
         raise YAMLParseError(f"error unmarshaling yaml, {err}") from err
+    finally:
+        comments = list(loader.collected_comments)
+        loader.dispose()
 
     docs = []
     for pydoc in pydocs:
@@ -205,8 +191,9 @@ def parse_documents(src: str) -> list[Node]:
         doc.content = [root]
         docs.append(doc)
 
-    comments = _scan_comments(src)
     if comments:
+        _fixup_full_line(src, comments)
+        comments.sort(key=lambda c: c["index"])
         _attach_comments(src, docs, comments)
 
     return docs
