@@ -1,0 +1,36 @@
+# Developer entry points for the generator itself
+# (analog of the reference's Makefile: build/test/func-test targets)
+
+PYTHON ?= python3
+
+.PHONY: all test test-unit test-func test-gpu bench smoke lint clean
+
+all: test
+
+test:
+	$(PYTHON) -m pytest tests/ -q -m "not gpu"
+
+# unit tiers only (marker engine, yamlast, domain model)
+test-unit:
+	$(PYTHON) -m pytest tests/test_lexer.py tests/test_marker_parser.py \
+		tests/test_yamlast.py tests/test_api_fields.py tests/test_rbac.py \
+		tests/test_companion_and_config.py tests/test_utils_and_misc.py -q
+
+# functional tiers (full init + create api over fixtures)
+test-func:
+	$(PYTHON) -m pytest tests/test_generate_standalone.py \
+		tests/test_generate_collection.py tests/test_generate_edge.py \
+		tests/test_reference_fixtures.py tests/test_api_upgrade.py -q
+
+test-gpu:
+	$(PYTHON) -m pytest tests/ -q -m gpu
+
+bench:
+	$(PYTHON) bench.py --steps 20 --warmup 3
+
+smoke:
+	$(PYTHON) __graft_entry__.py
+
+clean:
+	find . -name __pycache__ -type d -prune -exec rm -rf {} +
+	rm -rf .pytest_cache build dist *.egg-info
