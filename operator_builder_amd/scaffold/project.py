@@ -50,8 +50,9 @@ class Project:
             raise ProjectError(
                 f"no PROJECT file found at {path} - run `init` first"
             )
+        loader = getattr(yaml, "CSafeLoader", yaml.SafeLoader)
         with open(path, encoding="utf-8") as f:
-            raw = yaml.safe_load(f) or {}
+            raw = yaml.load(f, Loader=loader) or {}
 
         project = cls(
             domain=raw.get("domain", "") or "",

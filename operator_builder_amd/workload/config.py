@@ -138,9 +138,10 @@ def parse(config_path: str) -> Processor:
 
 
 def _parse_into(processor: Processor, validator: _InlineValidator) -> None:
+    loader = getattr(yaml, "CSafeLoader", yaml.SafeLoader)
     try:
         with open(processor.path, encoding="utf-8") as f:
-            raw_docs = list(yaml.safe_load_all(f))
+            raw_docs = list(yaml.load_all(f, Loader=loader))
     except OSError as err:
         raise ConfigError(
             f"error reading file {processor.path}; {err}"
