@@ -561,13 +561,9 @@ def cmd_generate_sub(ctx: Context, builder: Workload) -> File:
         inputs = "workloadFile"
 
     col = builder.get_collection()
-    col_import = ""
-    if builder.is_component():
-        col_import = (
-            f"\t{col.get_api_group()}{col.get_api_version()} "
-            f'"{ctx.repo}/apis/{col.get_api_group()}/'
-            f'{col.get_api_version()}"\n'
-        )
+    # NOTE: the reference's cmdGenerateSub template imports the collection
+    # API package for components but never references it (an unused import
+    # is a Go compile error), so no collection import is emitted here.
 
     options = []
     if use_collection_flag:
@@ -650,7 +646,7 @@ import (
 \tcmdgenerate "{ctx.repo}/cmd/{root.name}/commands/generate"
 
 \t// specific imports for workloads
-{col_import}\t{OB_IMPORTS_MARKER}
+\t{OB_IMPORTS_MARKER}
 )
 
 // New{kind}SubCommand creates a new command and adds it to its

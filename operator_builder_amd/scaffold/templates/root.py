@@ -1,6 +1,10 @@
 """Project-root templates: main.go, go.mod, Makefile, Dockerfile, README.
 
 Parity targets: reference templates/main.go:22-282 (Main + MainUpdater),
+with one correctness deviation: the reference's mainTemplate imports
+sigs.k8s.io/controller-runtime/pkg/controller without using it (an
+unused import is a Go compile error), so it is omitted here.
+Other parity targets:
 templates/gomod.go:23-66, templates/makefile.go, templates/dockerfile.go,
 templates/readme.go.
 """
@@ -53,7 +57,6 @@ import (
 \tutilruntime "k8s.io/apimachinery/pkg/util/runtime"
 \tclientgoscheme "k8s.io/client-go/kubernetes/scheme"
 \tctrl "sigs.k8s.io/controller-runtime"
-\t"sigs.k8s.io/controller-runtime/pkg/controller"
 \t"sigs.k8s.io/controller-runtime/pkg/log/zap"
 \t"sigs.k8s.io/controller-runtime/pkg/healthz"
 \t{IMPORT_MARKER}
