@@ -236,10 +236,24 @@ def build_parser() -> argparse.ArgumentParser:
     p_api = create_sub.add_parser(
         "api", help="Build a new API that can capture state for workloads"
     )
+    # kubebuilder-style bool flags: `--controller`, `--controller=false`
+    # (the reference's documented update workflow passes
+    # `--controller=false --resource --force`, docs/api-updates-upgrades.md)
+    def bool_flag(value):
+        if isinstance(value, bool):
+            return value
+        return value.lower() not in ("false", "0", "no")
+
     p_api.add_argument("--workload-config", default="")
-    p_api.add_argument("--controller", action="store_true")
-    p_api.add_argument("--resource", action="store_true")
-    p_api.add_argument("--force", action="store_true")
+    p_api.add_argument(
+        "--controller", nargs="?", const=True, default=True, type=bool_flag
+    )
+    p_api.add_argument(
+        "--resource", nargs="?", const=True, default=True, type=bool_flag
+    )
+    p_api.add_argument(
+        "--force", nargs="?", const=True, default=False, type=bool_flag
+    )
     p_api.add_argument("--directory", default=".")
     p_api.set_defaults(func=cmd_create_api)
 
