@@ -61,9 +61,16 @@ class Fragments:
 class Scaffold:
     def __init__(self, base_dir: str):
         self.base_dir = base_dir
+        self._made_dirs: set[str] = set()
 
     def _full(self, path: str) -> str:
         return os.path.join(self.base_dir, path)
+
+    def _ensure_dir(self, full_path: str) -> None:
+        directory = os.path.dirname(full_path) or "."
+        if directory not in self._made_dirs:
+            os.makedirs(directory, exist_ok=True)
+            self._made_dirs.add(directory)
 
     def execute(self, *items) -> None:
         for item in items:
@@ -85,7 +92,7 @@ class Scaffold:
                     f"failed to create {item.path}: file already exists"
                 )
 
-        os.makedirs(os.path.dirname(full) or ".", exist_ok=True)
+        self._ensure_dir(full)
         with open(full, "w", encoding="utf-8") as f:
             f.write(item.content)
 
@@ -97,7 +104,7 @@ class Scaffold:
                 raise ScaffoldError(
                     f"unable to insert fragments: {item.path} does not exist"
                 )
-            os.makedirs(os.path.dirname(full) or ".", exist_ok=True)
+            self._ensure_dir(full)
             with open(full, "w", encoding="utf-8") as f:
                 f.write(item.missing_file_content)
 

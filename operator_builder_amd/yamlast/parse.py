@@ -160,19 +160,102 @@ def _blank_line_between(src_lines, a: int, b: int) -> bool:
     return False
 
 
-def parse_documents(src: str) -> list[Node]:
-    """Parse a (possibly multi-document) YAML string into document Nodes
-    with comments attached."""
+_HAS_LIBYAML = hasattr(yaml, "CSafeLoader")
+
+
+def _compose_c(src: str):
+    """Compose with libyaml (≈3x faster than the pure-Python scanner) and
+    recover comments positionally: any '#' outside every scalar node's
+    source span begins a comment (block/flow scalars cover their own
+    content, so '#' inside values never false-positives)."""
+    loader = yaml.CSafeLoader(src)
+    try:
+        pydocs = []
+        while loader.check_node():
+            pydocs.append(loader.get_node())
+    finally:
+        loader.dispose()
+
+    covered: list[tuple[int, int]] = []
+
+    def collect_spans(node):
+        if isinstance(node, yaml.ScalarNode):
+            covered.append((node.start_mark.index, node.end_mark.index))
+        elif isinstance(node, (yaml.SequenceNode, yaml.MappingNode)):
+            children = (
+                node.value
+                if isinstance(node, yaml.SequenceNode)
+                else [n for pair in node.value for n in pair]
+            )
+            for child in children:
+                collect_spans(child)
+
+    seen_ids: set[int] = set()
+
+    def collect_once(node):
+        if id(node) in seen_ids:
+            return
+        seen_ids.add(id(node))
+        collect_spans(node)
+
+    for doc in pydocs:
+        collect_once(doc)
+
+    covered.sort()
+
+    comments = []
+    pos = 0
+    n = len(src)
+    ci = 0
+    while pos < n:
+        idx = src.find("#", pos)
+        if idx == -1:
+            break
+        # skip '#' covered by a scalar span
+        while ci < len(covered) and covered[ci][1] <= idx:
+            ci += 1
+        if ci < len(covered) and covered[ci][0] <= idx < covered[ci][1]:
+            pos = covered[ci][1]
+            continue
+        eol = src.find("\n", idx)
+        if eol == -1:
+            eol = n
+        comments.append(
+            dict(
+                index=idx,
+                line=src.count("\n", 0, idx),
+                column=idx - (src.rfind("\n", 0, idx) + 1),
+                text=src[idx:eol].rstrip(),
+                full_line=False,
+            )
+        )
+        pos = eol
+
+    return pydocs, comments
+
+
+def _compose_py(src: str):
     loader = _CommentLoader(src)
     try:
         pydocs = []
         while loader.check_node():
             pydocs.append(loader.get_node())
-    except yaml.YAMLError as err:
-        raise YAMLParseError(f"error unmarshaling yaml, {err}") from err
     finally:
         comments = list(loader.collected_comments)
         loader.dispose()
+    return pydocs, comments
+
+
+def parse_documents(src: str) -> list[Node]:
+    """Parse a (possibly multi-document) YAML string into document Nodes
+    with comments attached."""
+    try:
+        if _HAS_LIBYAML:
+            pydocs, comments = _compose_c(src)
+        else:
+            pydocs, comments = _compose_py(src)
+    except yaml.YAMLError as err:
+        raise YAMLParseError(f"error unmarshaling yaml, {err}") from err
 
     docs = []
     for pydoc in pydocs:
