@@ -31,8 +31,6 @@ from ..scaffold.scaffolder import (
 from ..scaffold.templates.base import APACHE2_BOILERPLATE
 from ..workload import config as workload_config
 from ..workload import kinds, subcommand
-from ..workload.companion import CLI as CompanionCLI
-from ..workload.manifests import Manifests
 
 
 class CLIError(Exception):

@@ -12,7 +12,7 @@ Parity target: reference internal/workload/v1/rbac:
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Any, Optional
+from typing import Any
 
 from ..utils import regular_plural
 

@@ -15,7 +15,7 @@ import yaml
 
 from . import kinds
 from .companion import CLI
-from .config import ConfigError, Processor
+from .config import Processor
 from .markers import MarkerCollection
 
 
