@@ -91,10 +91,10 @@ def cmd_init(args) -> int:
     )
     project.save(base_dir)
 
+    print("Adding workload scaffolding...")
+
     ctx = _build_context(base_dir, project, workload)
     scaffold_init(base_dir, ctx, workload)
-
-    print("Adding workload scaffolding...")
 
     return 0
 
@@ -117,6 +117,8 @@ def cmd_create_api(args) -> int:
 
     workload = processor.workload
 
+    print("Building API...")
+
     ctx = _build_context(base_dir, project, workload)
     scaffold_api(base_dir, ctx, workload)
 
@@ -125,8 +127,6 @@ def cmd_create_api(args) -> int:
         project.add_resource(resource_for_workload(ctx, w))
     project.plugin_config.workload_config_path = config_path
     project.save(base_dir)
-
-    print("Building API...")
 
     return 0
 
