@@ -1,0 +1,86 @@
+"""CLI process-level tests: the tool behaves as a real executable."""
+
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def run_cli(*args, cwd=None):
+    env = dict(os.environ, PYTHONPATH=REPO)
+    return subprocess.run(
+        [sys.executable, "-m", "operator_builder_amd.cli.main", *args],
+        capture_output=True,
+        text=True,
+        cwd=cwd or REPO,
+        env=env,
+        timeout=120,
+    )
+
+
+def test_help():
+    result = run_cli("--help")
+    assert result.returncode == 0
+    for command in ("init", "create", "init-config", "update", "version"):
+        assert command in result.stdout
+
+
+def test_version():
+    result = run_cli("version")
+    assert result.returncode == 0
+    assert "version" in result.stdout
+
+
+def test_completion_bash():
+    result = run_cli("completion", "bash")
+    assert result.returncode == 0
+    assert "complete -F" in result.stdout
+
+
+def test_init_config_stdout_is_valid_yaml():
+    import yaml
+
+    result = run_cli("init-config", "standalone")
+    assert result.returncode == 0
+    doc = yaml.safe_load(result.stdout)
+    assert doc["kind"] == "StandaloneWorkload"
+    assert doc["spec"]["api"]["domain"] == "acme.com"
+
+
+def test_missing_config_is_clean_error(tmp_path):
+    result = run_cli(
+        "init", "--workload-config", "nope.yaml", cwd=str(tmp_path)
+    )
+    assert result.returncode == 1
+    assert "FATAL" in result.stderr
+
+
+def test_create_api_without_project_is_clean_error(tmp_path):
+    result = run_cli("create", "api", cwd=str(tmp_path))
+    assert result.returncode == 1
+    assert "PROJECT" in result.stderr
+
+
+def test_full_generation_via_subprocess(tmp_path):
+    import shutil
+
+    workdir = tmp_path / "proj"
+    workdir.mkdir()
+    shutil.copytree(
+        os.path.join(REPO, "tests", "fixtures", "standalone"),
+        workdir / ".workloadConfig",
+    )
+    result = run_cli(
+        "init",
+        "--workload-config",
+        ".workloadConfig/workload.yaml",
+        "--repo",
+        "github.com/acme/app",
+        cwd=str(workdir),
+    )
+    assert result.returncode == 0, result.stderr
+    result = run_cli("create", "api", cwd=str(workdir))
+    assert result.returncode == 0, result.stderr
+    assert (workdir / "apis" / "apps" / "v1alpha1").is_dir()
