@@ -1,0 +1,97 @@
+"""Conformance: validation behavior over the reference's own
+test/configs fixtures (valid + invalid workload configs). Skipped when
+the reference checkout is absent."""
+
+import glob
+import os
+
+import pytest
+
+from operator_builder_amd.workload import config
+
+CONFIGS = "/root/reference/test/configs"
+
+pytestmark = pytest.mark.skipif(
+    not os.path.isdir(CONFIGS), reason="reference checkout not available"
+)
+
+
+def test_standalone_valid_parses():
+    processor = config.parse(os.path.join(CONFIGS, "standalone", "valid.yaml"))
+    assert processor.workload.is_standalone()
+    assert processor.workload.get_api_kind() == "MyApp"
+
+
+@pytest.mark.parametrize(
+    "missing", ["domain", "group", "kind", "name", "version"]
+)
+def test_standalone_invalid_missing_fields(missing):
+    path = os.path.join(
+        CONFIGS, "standalone", f"invalid-missing-{missing}.yaml"
+    )
+    with pytest.raises(config.ConfigError):
+        config.parse(path)
+
+
+def test_collection_valid_parses_with_components():
+    processor = config.parse(os.path.join(CONFIGS, "collection", "valid.yaml"))
+    assert processor.workload.is_collection()
+    assert len(processor.children) >= 1
+
+
+@pytest.mark.parametrize(
+    "case",
+    [
+        "invalid-missing-domain",
+        "invalid-missing-group",
+        "invalid-missing-kind",
+        "invalid-missing-name",
+        "invalid-missing-version",
+        "invalid-missing-dependencies",
+        "invalid-overlapping-kinds",
+        "invalid-overlapping-names",
+    ],
+)
+def test_collection_invalid_cases(case):
+    path = os.path.join(CONFIGS, "collection", f"{case}.yaml")
+    with pytest.raises(config.ConfigError):
+        config.parse(path)
+
+
+def test_collection_overlap_errors_are_specific():
+    with pytest.raises(config.ConfigError, match="unique"):
+        config.parse(
+            os.path.join(CONFIGS, "collection", "invalid-overlapping-names.yaml")
+        )
+    with pytest.raises(config.ConfigError, match="already exists in group"):
+        config.parse(
+            os.path.join(CONFIGS, "collection", "invalid-overlapping-kinds.yaml")
+        )
+    with pytest.raises(config.ConfigError, match="dependencies"):
+        config.parse(
+            os.path.join(
+                CONFIGS, "collection", "invalid-missing-dependencies.yaml"
+            )
+        )
+
+
+@pytest.mark.parametrize(
+    "case", glob.glob(os.path.join(CONFIGS, "component", "invalid-*.yaml"))
+    if os.path.isdir(CONFIGS)
+    else [],
+)
+def test_component_invalid_cases(case):
+    with pytest.raises(config.ConfigError):
+        config.parse(case)
+
+
+def test_component_alone_requires_collection():
+    # a valid component parsed as the top-level config is rejected
+    # (reference parse.go:50-60)
+    with pytest.raises(config.ConfigError, match="WorkloadCollection"):
+        config.parse(os.path.join(CONFIGS, "component", "valid.yaml"))
+
+
+def test_unknown_kind_rejected():
+    with pytest.raises(config.ConfigError, match="unrecognized"):
+        config.parse(os.path.join(CONFIGS, "invalid-type.yaml"))
