@@ -194,10 +194,42 @@ _operator_builder_completions() {
 complete -F _operator_builder_completions operator-builder
 """
 
+COMPLETION_ZSH = """#compdef operator-builder
+# zsh completion for operator-builder
+_operator_builder() {
+    local -a commands
+    commands=(
+        'init:Initialize a new operator project'
+        'create:Scaffold into the project'
+        'init-config:Initialize a workload configuration'
+        'update:Update an existing project'
+        'version:Print version information'
+        'completion:Generate shell completion scripts'
+    )
+    _describe 'command' commands
+}
+_operator_builder "$@"
+"""
+
+COMPLETION_FISH = """# fish completion for operator-builder
+complete -c operator-builder -f
+complete -c operator-builder -n __fish_use_subcommand -a init -d 'Initialize a new operator project'
+complete -c operator-builder -n __fish_use_subcommand -a create -d 'Scaffold into the project'
+complete -c operator-builder -n __fish_use_subcommand -a init-config -d 'Initialize a workload configuration'
+complete -c operator-builder -n __fish_use_subcommand -a update -d 'Update an existing project'
+complete -c operator-builder -n __fish_use_subcommand -a version -d 'Print version information'
+complete -c operator-builder -n __fish_use_subcommand -a completion -d 'Generate shell completion scripts'
+"""
+
 
 def cmd_completion(args) -> int:
-    if args.shell == "bash":
-        print(COMPLETION_BASH)
+    scripts = {
+        "bash": COMPLETION_BASH,
+        "zsh": COMPLETION_ZSH,
+        "fish": COMPLETION_FISH,
+    }
+    if args.shell in scripts:
+        print(scripts[args.shell])
         return 0
     raise CLIError(f"unsupported shell: {args.shell}")
 
@@ -286,7 +318,7 @@ def build_parser() -> argparse.ArgumentParser:
     p_completion = sub.add_parser(
         "completion", help="Generate shell completion scripts"
     )
-    p_completion.add_argument("shell", choices=["bash"])
+    p_completion.add_argument("shell", choices=["bash", "zsh", "fish"])
     p_completion.set_defaults(func=cmd_completion)
 
     return parser
