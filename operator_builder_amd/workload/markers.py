@@ -52,15 +52,11 @@ class FieldType(enum.Enum):
 
     @classmethod
     def unmarshal(cls, raw: str) -> "FieldType":
-        try:
-            ft = cls(raw)
-        except ValueError:
-            raise MarkerError(
-                f"unable to parse field, {raw} into FieldType"
-            ) from None
-        if ft == cls.UNKNOWN:
-            raise MarkerError(f"unable to parse field, {raw} into FieldType")
-        return ft
+        # only scalar types are accepted from marker args; struct fields
+        # arise implicitly from dotted paths (reference field_types.go:29-47)
+        if raw in ("string", "int", "bool"):
+            return cls(raw)
+        raise MarkerError(f"unable to parse field, {raw} into FieldType")
 
 
 @dataclass
