@@ -203,6 +203,13 @@ def _compose_c(src: str):
 
     covered.sort()
 
+    import bisect
+
+    line_starts = [0]
+    for i, ch in enumerate(src):
+        if ch == "\n":
+            line_starts.append(i + 1)
+
     comments = []
     pos = 0
     n = len(src)
@@ -220,11 +227,12 @@ def _compose_c(src: str):
         eol = src.find("\n", idx)
         if eol == -1:
             eol = n
+        line = bisect.bisect_right(line_starts, idx) - 1
         comments.append(
             dict(
                 index=idx,
-                line=src.count("\n", 0, idx),
-                column=idx - (src.rfind("\n", 0, idx) + 1),
+                line=line,
+                column=idx - line_starts[line],
                 text=src[idx:eol].rstrip(),
                 full_line=False,
             )
@@ -283,10 +291,13 @@ def parse_documents(src: str) -> list[Node]:
 
 
 def _attach_comments(src: str, docs: list[Node], comments: list[dict]):
+    import bisect
+
     src_lines = src.split("\n")
 
-    # collect candidate nodes: all nodes of all documents, with depth so
-    # "innermost at a position" is resolvable
+    # collect candidate nodes once, with depth for innermost-at-position
+    # resolution; all lookups below are indexed (bisect), keeping this
+    # linear-ish even for very large marker-dense manifests
     nodes: list[tuple[Node, int]] = []
 
     def collect(node: Node, depth: int):
@@ -301,20 +312,42 @@ def _attach_comments(src: str, docs: list[Node], comments: list[dict]):
     if not nodes:
         return
 
-    scalars = [n for n, _ in nodes if n.kind == SCALAR]
+    # scalars ending on each line, for trailing-comment attachment
+    scalars_by_end_line: dict[int, list[Node]] = {}
+    for n, _ in nodes:
+        if n.kind == SCALAR:
+            scalars_by_end_line.setdefault(n.end_line, []).append(n)
+    nodes_by_end_line: dict[int, list[Node]] = {}
+    for n, _ in nodes:
+        nodes_by_end_line.setdefault(n.end_line, []).append(n)
+
+    # all nodes sorted by (start index, -depth): the first entry past an
+    # offset is the innermost node at the next content position
+    starts = sorted(
+        ((n.index, -depth, id(n), n) for n, depth in nodes),
+        key=lambda t: (t[0], t[1], t[2]),
+    )
+    start_keys = [t[0] for t in starts]
+
+    # scalars sorted by end index, for previous-node (foot) attachment
+    scalar_ends = sorted(
+        ((n.end_index, id(n), n) for n, _ in nodes if n.kind == SCALAR),
+        key=lambda t: (t[0], t[1]),
+    )
+    scalar_end_keys = [t[0] for t in scalar_ends]
 
     # ---- line comments -------------------------------------------------
-    trailing = [c for c in comments if not c["full_line"]]
-    for c in trailing:
+    for c in comments:
+        if c["full_line"]:
+            continue
         best = None
-        for n in scalars:
-            if n.end_line == c["line"] and n.end_index <= c["index"]:
+        for n in scalars_by_end_line.get(c["line"], []):
+            if n.end_index <= c["index"]:
                 if best is None or n.end_index > best.end_index:
                     best = n
         if best is None:
-            # e.g. a comment after a flow collection or on a key line
-            for n, _ in nodes:
-                if n.end_line == c["line"] and n.end_index <= c["index"]:
+            for n in nodes_by_end_line.get(c["line"], []):
+                if n.end_index <= c["index"]:
                     if best is None or n.end_index > best.end_index:
                         best = n
         if best is not None:
@@ -337,25 +370,13 @@ def _attach_comments(src: str, docs: list[Node], comments: list[dict]):
         first, last = block[0], block[-1]
         text = "\n".join(c["text"] for c in block)
 
-        # next node strictly after the block
-        nxt = None
-        nxt_depth = -1
-        for n, depth in nodes:
-            if n.index > last["index"]:
-                if (
-                    nxt is None
-                    or n.index < nxt.index
-                    or (n.index == nxt.index and depth > nxt_depth)
-                ):
-                    nxt = n
-                    nxt_depth = depth
+        # next (innermost) node strictly after the block
+        pos = bisect.bisect_right(start_keys, last["index"])
+        nxt = starts[pos][3] if pos < len(starts) else None
 
-        # previous node ending before the block
-        prev = None
-        for n, _ in nodes:
-            if n.kind == SCALAR and n.end_index <= first["index"]:
-                if prev is None or n.end_index > prev.end_index:
-                    prev = n
+        # previous scalar ending before the block
+        pos = bisect.bisect_right(scalar_end_keys, first["index"])
+        prev = scalar_ends[pos - 1][2] if pos > 0 else None
 
         attach_head = nxt is not None and not _blank_line_between(
             src_lines, last["line"], nxt.line
@@ -364,10 +385,12 @@ def _attach_comments(src: str, docs: list[Node], comments: list[dict]):
         if attach_head:
             if nxt.kind != SCALAR:
                 # prefer the innermost scalar starting at the same index
-                for n, _ in nodes:
-                    if n.index == nxt.index and n.kind == SCALAR:
-                        nxt = n
+                pos = bisect.bisect_left(start_keys, nxt.index)
+                while pos < len(starts) and starts[pos][0] == nxt.index:
+                    if starts[pos][3].kind == SCALAR:
+                        nxt = starts[pos][3]
                         break
+                    pos += 1
             if nxt.head_comment:
                 nxt.head_comment += "\n" + text
             else:
