@@ -130,3 +130,26 @@ def test_go_files_brace_balanced(tmp_path):
             rel = os.path.relpath(path, workdir)
             assert depth == 0, f"unbalanced braces in {rel}"
             assert in_str is None, f"unterminated string in {rel}"
+
+
+@pytest.mark.parametrize("fixture", ["standalone", "collection"])
+def test_generated_yaml_files_are_valid(tmp_path, fixture):
+    import yaml as pyyaml
+
+    workdir = tmp_path / "gen"
+    workdir.mkdir()
+    generate(str(workdir), fixture, "github.com/acme/app")
+
+    checked = 0
+    for root, dirs, files in os.walk(workdir):
+        if ".workloadConfig" in root:
+            continue
+        for name in files:
+            if not name.endswith((".yaml", ".yml")):
+                continue
+            path = os.path.join(root, name)
+            with open(path, encoding="utf-8") as f:
+                docs = list(pyyaml.safe_load_all(f))
+            assert docs, f"empty yaml at {path}"
+            checked += 1
+    assert checked >= 20  # config tree + samples
