@@ -140,3 +140,35 @@ def test_cluster_scoped_collection_sample(project):
         project, "config/samples/platforms_v1alpha1_cloudplatform.yaml"
     )
     assert "namespace: default" not in sample.split("spec:")[0]
+
+
+def test_component_generate_subcommand_flags(project):
+    content = read(project, "cmd/platformctl/commands/generate/apps/webapp.go")
+    # components take both workload and collection manifest flags
+    assert "UseCollectionManifest: true," in content
+    assert 'CollectionKind:        "CloudPlatform",' in content
+    assert "UseWorkloadManifest:   true," in content
+    assert 'WorkloadKind:          "WebApp",' in content
+    # component generate funcs take two manifests
+    assert "type generateFunc func([]byte, []byte) ([]client.Object, error)" in content
+    assert '"v1alpha1": v1alpha1webapp.GenerateForCLI,' in content
+
+
+def test_collection_generate_subcommand_flags(project):
+    content = read(
+        project, "cmd/platformctl/commands/generate/platforms/cloudplatform.go"
+    )
+    # the collection takes only its own (collection) manifest
+    assert "UseCollectionManifest: true," in content
+    assert 'CollectionKind:        "CloudPlatform",' in content
+    assert "UseWorkloadManifest" not in content
+    assert "type generateFunc func([]byte) ([]client.Object, error)" in content
+
+
+def test_component_resources_generate_for_cli_signature(project):
+    content = read(project, "apis/apps/v1alpha1/webapp/resources.go")
+    assert (
+        "func GenerateForCLI(workloadFile []byte,collectionFile []byte,)"
+        in content
+    )
+    assert "return Generate(workloadObj, collectionObj)" in content
