@@ -1,0 +1,93 @@
+//go:build e2e_test
+// +build e2e_test
+
+/*
+Copyright 2021.
+
+Licensed under the Apache License, Version 2.0 (the "License");
+you may not use this file except in compliance with the License.
+You may obtain a copy of the License at
+
+    http://www.apache.org/licenses/LICENSE-2.0
+
+Unless required by applicable law or agreed to in writing, software
+distributed under the License is distributed on an "AS IS" BASIS,
+WITHOUT WARRANTIES OR CONDITIONS OF ANY KIND, either express or implied.
+See the License for the specific language governing permissions and
+limitations under the License.
+*/
+
+package e2e_test
+
+import (
+	"fmt"
+	"os"
+
+	"github.com/stretchr/testify/require"
+
+	"k8s.io/apimachinery/pkg/apis/meta/v1/unstructured"
+
+	appsv1alpha1 "github.com/acme/bookstore/apis/apps/v1alpha1"
+	"github.com/acme/bookstore/apis/apps/v1alpha1/bookstore"
+)
+
+//
+// appsv1alpha1BookStore tests
+//
+func appsv1alpha1BookStoreChildrenFuncs(tester *E2ETest) error {
+	if len(bookstore.CreateFuncs) == 0 {
+		return nil
+	}
+
+	workload, err := bookstore.ConvertWorkload(tester.workload)
+	if err != nil {
+		return fmt.Errorf("error in workload conversion; %w", err)
+	}
+
+	resourceObjects, err := bookstore.Generate(*workload)
+	if err != nil {
+		return fmt.Errorf("unable to create objects in memory; %w", err)
+	}
+
+	tester.children = resourceObjects
+
+	return nil
+}
+
+func appsv1alpha1BookStoreNewHarness(namespace string) *E2ETest {
+	return &E2ETest{
+		namespace:          namespace,
+		unstructured:       &unstructured.Unstructured{},
+		workload:           &appsv1alpha1.BookStore{},
+		sampleManifestFile: "../../config/samples/apps_v1alpha1_bookstore.yaml",
+		getChildrenFunc:    appsv1alpha1BookStoreChildrenFuncs,
+		logSyntax:          "controllers.apps.BookStore",
+	}
+}
+
+func (tester *E2ETest) appsv1alpha1BookStoreTest(testSuite *E2EComponentTestSuite) {
+	testSuite.suiteConfig.tests = append(testSuite.suiteConfig.tests, tester)
+	tester.suiteConfig = &testSuite.suiteConfig
+	require.NoErrorf(testSuite.T(), tester.setup(), "failed to setup test")
+
+	// create the custom resource and wait for its children to be ready
+	require.NoErrorf(testSuite.T(), testCreateCustomResource(tester), "failed to create custom resource")
+
+	// delete a whitelisted child and wait for the controller to restore it
+	require.NoErrorf(testSuite.T(), testDeleteChildResource(tester), "failed to reconcile deletion of a child resource")
+
+	// verify the controller logged no errors for this workload
+	if os.Getenv("DEPLOY_IN_CLUSTER") == "true" {
+		require.NoErrorf(testSuite.T(), testControllerLogsNoErrors(tester.suiteConfig, tester.logSyntax), "found errors in controller logs")
+	}
+}
+
+func (testSuite *E2EComponentTestSuite) Test_appsv1alpha1BookStore() {
+	tester := appsv1alpha1BookStoreNewHarness("test-apps-v1alpha1-bookstore")
+	tester.appsv1alpha1BookStoreTest(testSuite)
+}
+
+func (testSuite *E2EComponentTestSuite) Test_appsv1alpha1BookStoreMulti() {
+	tester := appsv1alpha1BookStoreNewHarness("test-apps-v1alpha1-bookstore-2")
+	tester.appsv1alpha1BookStoreTest(testSuite)
+}
