@@ -115,8 +115,10 @@ def _scaffold_workload(
     # e2e workload test
     scaffold.execute(e2e_tpl.workload_test(workload_ctx, workload))
 
-    # companion CLI subcommands
-    if ctx.cli_root_command_name:
+    # companion CLI subcommands — only for workloads that belong to a
+    # companion CLI (a later standalone added to a project that has one
+    # may itself define none)
+    if ctx.cli_root_command_name and workload.get_root_command().name:
         _scaffold_cli(scaffold, workload_ctx, workload)
 
     # recurse into collection components
