@@ -82,6 +82,14 @@ class Inspector:
     def _inspect_comments(self, *nodes: Node) -> list[YAMLResult]:
         markers: list[Result] = []
         for node in nodes:
+            # fast path: most nodes carry no comments (and no comment
+            # without a '+' can contain a marker) — skip the lexer run
+            if not (
+                "+" in node.head_comment
+                or "+" in node.line_comment
+                or "+" in node.foot_comment
+            ):
+                continue
             text = "\n".join(
                 (node.head_comment, node.line_comment, node.foot_comment)
             )
