@@ -242,3 +242,28 @@ class TestLicense:
         assert (tmp_path / "sub" / "y.go").read_text().startswith(
             "// licensed"
         )
+
+
+class TestInitConfigRoundTrip:
+    def test_sample_config_feeds_init(self, tmp_path, monkeypatch):
+        """init-config output is itself a valid workload config."""
+        cfg = tmp_path / "workload.yaml"
+        assert main(["init-config", "standalone", "--path", str(cfg)]) == 0
+
+        workdir = tmp_path / "proj"
+        workdir.mkdir()
+        monkeypatch.chdir(workdir)
+        assert (
+            main(
+                [
+                    "init",
+                    "--workload-config",
+                    str(cfg),
+                    "--repo",
+                    "github.com/acme/sample",
+                ]
+            )
+            == 0
+        )
+        assert os.path.exists("PROJECT")
+        assert os.path.exists("main.go")
