@@ -194,3 +194,14 @@ data:
     api.add_field("shared", FieldType.STRING, None, "1", False)
     with pytest.raises(APIFieldError):
         api.add_field("shared", FieldType.INT, None, 2, False)
+
+
+def test_boolean_like_field_name_rejected():
+    """A field named 'true' lexes as a boolean literal in the marker args
+    and fails arg typing — faithful to the reference's lexer ordering
+    (lexBooleanLiteral before lexNakedStringLiteral, state.go:160-176)."""
+    with pytest.raises(Exception, match="incorrect type"):
+        inspect_for_yaml(
+            'x: "1"  # +operator-builder:field:name=true,type=string\n',
+            MarkerType.FIELD,
+        )
