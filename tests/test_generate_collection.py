@@ -172,3 +172,25 @@ def test_component_resources_generate_for_cli_signature(project):
         in content
     )
     assert "return Generate(workloadObj, collectionObj)" in content
+
+
+def test_collection_create_api_idempotent(project):
+    assert main(["create", "api"]) == 0
+    assert main(["create", "api"]) == 0
+
+    main_go = read(project, "main.go")
+    for fragment in (
+        "platformscontrollers.NewCloudPlatformReconciler(mgr),",
+        "appscontrollers.NewWebAppReconciler(mgr),",
+        "datacontrollers.NewDataStoreReconciler(mgr),",
+    ):
+        assert main_go.count(fragment) == 1, fragment
+
+    root_cmd = read(project, "cmd/platformctl/commands/root.go")
+    assert root_cmd.count("initapps.NewWebAppSubCommand(parentCommand)") == 1
+
+    crd = read(project, "config/crd/kustomization.yaml")
+    assert crd.count("- bases/apps.example.com_webapps.yaml") == 1
+
+    kind_file = read(project, "apis/apps/webapp.go")
+    assert kind_file.count("v1alpha1apps.GroupVersion,") == 1
