@@ -76,3 +76,51 @@ def test_generated_tree_matches_golden(generated):
         f"content diverged from golden snapshot in: {diverged[:10]} "
         "(see module docstring to regenerate intentionally)"
     )
+
+
+def test_collection_tree_matches_hash_manifest(tmp_path, monkeypatch):
+    """The collection fixture's 89-file tree is locked by sha256 manifest
+    (tests/golden/collection.sha256.json). Regenerate it with the loop in
+    this test (dump `actual`) after intentional template changes."""
+    import hashlib
+    import json
+
+    manifest_path = os.path.join(
+        os.path.dirname(__file__), "golden", "collection.sha256.json"
+    )
+    with open(manifest_path) as f:
+        golden = json.load(f)
+
+    workdir = tmp_path / "collection"
+    workdir.mkdir()
+    shutil.copytree(
+        os.path.join(FIXTURES, "collection"), workdir / ".workloadConfig"
+    )
+    monkeypatch.chdir(workdir)
+    assert (
+        main(
+            [
+                "init",
+                "--workload-config",
+                ".workloadConfig/workload.yaml",
+                "--repo",
+                "github.com/acme/platform",
+            ]
+        )
+        == 0
+    )
+    assert main(["create", "api"]) == 0
+
+    actual = {}
+    for root, dirs, files in os.walk(workdir):
+        if ".workloadConfig" in root:
+            continue
+        for name in sorted(files):
+            path = os.path.join(root, name)
+            rel = os.path.relpath(path, workdir)
+            with open(path, "rb") as f:
+                actual[rel] = hashlib.sha256(f.read()).hexdigest()
+
+    assert sorted(actual) == sorted(golden)
+    diverged = [rel for rel in golden if actual[rel] != golden[rel]]
+    assert diverged == [], diverged[:10]
