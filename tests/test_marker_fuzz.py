@@ -79,7 +79,7 @@ def manifests(draw):
     return "\n".join(lines) + "\n", fields
 
 
-@settings(max_examples=120, deadline=None)
+@settings(max_examples=120, deadline=None, derandomize=True)
 @given(manifests())
 def test_pipeline_invariants(case):
     src, fields = case
