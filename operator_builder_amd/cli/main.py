@@ -180,6 +180,37 @@ def cmd_update_license(args) -> int:
     return 0
 
 
+def cmd_validate(args) -> int:
+    """Dry-run validation (extension beyond the reference's commands):
+    parse the workload config, load the manifests, and run the full
+    marker-processing pipeline without writing any files."""
+    processor = workload_config.parse(args.workload_config)
+    subcommand.create_api(processor)
+
+    workloads = processor.get_workloads()
+    total_children = 0
+    total_markers = 0
+    for w in workloads:
+        spec = w.spec
+        children = sum(len(m.child_resources) for m in spec.manifests)
+        markers = len(spec.field_markers) + len(spec.collection_field_markers)
+        total_children += children
+        total_markers += markers
+        print(
+            f"ok: {w.get_workload_kind()} {w.get_name()} "
+            f"(kind={w.get_api_kind()}, group={w.get_api_group()}, "
+            f"version={w.get_api_version()}): "
+            f"{len(spec.manifests)} manifest file(s), "
+            f"{children} child resource(s), {markers} marker(s)"
+        )
+
+    print(
+        f"valid: {len(workloads)} workload(s), {total_children} child "
+        f"resource(s), {total_markers} marker(s)"
+    )
+    return 0
+
+
 def cmd_version(args) -> int:
     print(f"operator-builder-amd version {__version__}")
     return 0
@@ -309,6 +340,14 @@ def build_parser() -> argparse.ArgumentParser:
     p_lic.add_argument("--project-license", "-p", default="")
     p_lic.add_argument("--source-header-license", "-s", default="")
     p_lic.set_defaults(func=cmd_update_license)
+
+    # validate (extension)
+    p_validate = sub.add_parser(
+        "validate",
+        help="Validate a workload config and its manifests without scaffolding",
+    )
+    p_validate.add_argument("--workload-config", required=True)
+    p_validate.set_defaults(func=cmd_validate)
 
     # version
     p_version = sub.add_parser("version", help="Print version information")

@@ -84,3 +84,54 @@ def test_full_generation_via_subprocess(tmp_path):
     result = run_cli("create", "api", cwd=str(workdir))
     assert result.returncode == 0, result.stderr
     assert (workdir / "apis" / "apps" / "v1alpha1").is_dir()
+
+
+def test_validate_command(tmp_path):
+    import shutil
+
+    workdir = tmp_path / "v"
+    workdir.mkdir()
+    shutil.copytree(
+        os.path.join(REPO, "tests", "fixtures", "collection"),
+        workdir / ".workloadConfig",
+    )
+    result = run_cli(
+        "validate",
+        "--workload-config",
+        ".workloadConfig/workload.yaml",
+        cwd=str(workdir),
+    )
+    assert result.returncode == 0, result.stderr
+    assert "valid: 3 workload(s)" in result.stdout
+    # nothing scaffolded
+    assert not (workdir / "PROJECT").exists()
+
+
+def test_validate_command_rejects_bad_marker(tmp_path):
+    cfg = tmp_path / ".workloadConfig"
+    cfg.mkdir()
+    (cfg / "workload.yaml").write_text(
+        """name: app
+kind: StandaloneWorkload
+spec:
+  api:
+    domain: example.com
+    group: apps
+    version: v1
+    kind: App
+  resources:
+  - r.yaml
+"""
+    )
+    (cfg / "r.yaml").write_text(
+        'kind: ConfigMap\napiVersion: v1\nmetadata:\n  name: c\ndata:\n'
+        '  x: "1"  # +operator-builder:field:name=x,type=bogus\n'
+    )
+    result = run_cli(
+        "validate",
+        "--workload-config",
+        str(cfg / "workload.yaml"),
+        cwd=str(tmp_path),
+    )
+    assert result.returncode == 1
+    assert "FATAL" in result.stderr
