@@ -91,7 +91,14 @@ def run(args) -> None:
     rank = int(os.environ.get("RANK", "0"))
     distributed = world_size > 1
 
-    use_cuda = torch.cuda.is_available()
+    # use the GPU/RCCL path only when every local rank has its own
+    # device (a 2-process run on a 1-GPU box falls back to gloo)
+    local_world = int(
+        os.environ.get("LOCAL_WORLD_SIZE", os.environ.get("WORLD_SIZE", "1"))
+    )
+    use_cuda = (
+        torch.cuda.is_available() and torch.cuda.device_count() >= local_world
+    )
 
     if distributed:
         import torch.distributed as dist
