@@ -130,3 +130,67 @@ spec:
     content = spec.manifests[0].content
     assert "controlled by field: x" in content
     assert "the x value" in content
+
+
+def test_manifest_evolution_regenerates(tmp_path, monkeypatch):
+    """Adding a marker to a manifest and re-running create api surfaces
+    the new field (docs/api-updates-upgrades.md update workflow)."""
+    root = tmp_path / "proj"
+    cfg = root / ".workloadConfig"
+    write(
+        str(cfg / "workload.yaml"),
+        """name: app
+kind: StandaloneWorkload
+spec:
+  api:
+    domain: example.com
+    group: apps
+    version: v1
+    kind: App
+  resources:
+  - r.yaml
+""",
+    )
+    write(
+        str(cfg / "r.yaml"),
+        'kind: ConfigMap\napiVersion: v1\nmetadata:\n  name: c\ndata:\n'
+        '  a: "1"  # +operator-builder:field:name=alpha,type=string\n'
+        '  b: "2"\n',
+    )
+    monkeypatch.chdir(root)
+    assert (
+        main(
+            [
+                "init",
+                "--workload-config",
+                ".workloadConfig/workload.yaml",
+                "--repo",
+                "github.com/acme/app",
+            ]
+        )
+        == 0
+    )
+    assert main(["create", "api"]) == 0
+
+    types = read(root, "apis/apps/v1/app_types.go")
+    assert "Alpha string" in types
+    assert "Beta" not in types
+
+    # evolve the manifest: mark the second value too
+    write(
+        str(cfg / "r.yaml"),
+        'kind: ConfigMap\napiVersion: v1\nmetadata:\n  name: c\ndata:\n'
+        '  a: "1"  # +operator-builder:field:name=alpha,type=string\n'
+        '  b: "2"  # +operator-builder:field:name=beta,type=string,default="2"\n',
+    )
+    assert main(["create", "api"]) == 0
+
+    types = read(root, "apis/apps/v1/app_types.go")
+    assert "Alpha string" in types
+    assert "Beta string" in types
+
+    definition = read(root, "apis/apps/v1/app/r.go")
+    assert "parent.Spec.Beta" in definition
+
+    sample = read(root, "config/samples/apps_v1_app.yaml")
+    assert "beta:" in sample
