@@ -18,12 +18,7 @@ for mod, name in [
     fn = getattr(mod, name)
     inner = fn.hypothesis.inner_test
     # rebuild with a fresh randomized settings object
-    if mod is t1 and name == "test_roundtrip_preserves_value":
-        strat = t1.documents
-    elif mod is t1:
-        strat = t1.documents
-    else:
-        strat = t2.manifests()
+    strat = t1.documents if mod is t1 else t2.manifests()
     rebuilt = settings(max_examples=budget, deadline=None)(given(strat)(inner))
     rebuilt()
     print(f"ok: {name} x{budget}")
