@@ -102,3 +102,41 @@ def test_windows_line_endings():
     src = "a: 1\r\nb: 2  # +m:x:y\r\n"
     docs = parse_documents(src)
     assert to_plain(docs[0]) == {"a": 1, "b": 2}
+
+
+def test_unicode_values_roundtrip():
+    src = 'labels:\n  app: "caf\u00e9-\u4e2d\u6587"\n  emoji: "\u2728"\n'
+    out = roundtrip(src)
+    assert sem_equal(src, out)
+
+
+def test_comment_only_and_empty_documents():
+    src = "# just a comment\n---\nreal: 1\n---\n# trailing only\n"
+    docs = parse_documents(src)
+    # empty documents are dropped (reference decodes only real docs)
+    plains = [to_plain(d) for d in docs]
+    assert {"real": 1} in plains
+
+
+def test_empty_manifest_file_processing(tmp_path):
+    from operator_builder_amd.workload import config, subcommand
+
+    cfg = tmp_path / ".workloadConfig"
+    cfg.mkdir()
+    (cfg / "workload.yaml").write_text(
+        """name: app
+kind: StandaloneWorkload
+spec:
+  api:
+    domain: example.com
+    group: apps
+    version: v1
+    kind: App
+  resources:
+  - empty.yaml
+"""
+    )
+    (cfg / "empty.yaml").write_text("# nothing but a comment\n")
+    processor = config.parse(str(cfg / "workload.yaml"))
+    subcommand.create_api(processor)
+    assert processor.workload.spec.manifests[0].child_resources == []
