@@ -82,6 +82,14 @@ class InitConfigOptions:
     workload_config: kinds.Workload
 
 
+class _IndentedDumper(yaml.SafeDumper):
+    """Indent block sequence items under their key, matching the
+    reference's yaml.v3 encoder output (init_config.go:55-57)."""
+
+    def increase_indent(self, flow=False, indentless=False):
+        return super().increase_indent(flow, False)
+
+
 class InitConfigError(Exception):
     pass
 
@@ -130,7 +138,14 @@ def _marshal_config(workload: kinds.Workload) -> str:
     out["spec"] = spec
 
     buf = io.StringIO()
-    yaml.safe_dump(out, buf, sort_keys=True, default_flow_style=False, indent=2)
+    yaml.dump(
+        out,
+        buf,
+        Dumper=_IndentedDumper,
+        sort_keys=True,
+        default_flow_style=False,
+        indent=2,
+    )
     return buf.getvalue()
 
 
