@@ -95,3 +95,15 @@ def test_component_alone_requires_collection():
 def test_unknown_kind_rejected():
     with pytest.raises(config.ConfigError, match="unrecognized"):
         config.parse(os.path.join(CONFIGS, "invalid-type.yaml"))
+
+
+def test_init_config_standalone_byte_identical(capsys):
+    """`init-config standalone` output is byte-identical to the sample
+    the reference committed at test/configs/standalone/valid.yaml."""
+    from operator_builder_amd.cli.main import main
+
+    assert main(["init-config", "standalone"]) == 0
+    out = capsys.readouterr().out
+    with open(os.path.join(CONFIGS, "standalone", "valid.yaml")) as f:
+        golden = f.read()
+    assert out == golden
