@@ -190,3 +190,42 @@ class TestTransform:
         )
         key = docs[0].root.content[0]
         assert key.head_comment == "# controlled by field: x"
+
+
+class TestMultiMarkerBlockScalar:
+    """Scenario from the reference's docs/markers.md: several replace
+    markers attached to one block-scalar value, each splicing a regex
+    match inside the literal text."""
+
+    SRC = """kind: ConfigMap
+apiVersion: v1
+metadata:
+  name: contour-configmap
+data:
+  # +operator-builder:field:name=configOption,default=myoption,type=string,replace="configuration2"
+  # +operator-builder:field:name=yamlType,default=myoption,type=string,replace="multi.*yaml"
+  config.yaml: |
+    someoption: configuration2
+    anotheroption: configuration1
+    justtesting: multistringyaml
+"""
+
+    def test_both_markers_splice(self):
+        docs, results = inspect_for_yaml(self.SRC, MarkerType.FIELD)
+        assert len(results) == 2
+
+        value = docs[0].root.get("data").get("config.yaml")
+        assert "!!start parent.Spec.ConfigOption !!end" in value.value
+        assert "!!start parent.Spec.YamlType !!end" in value.value
+        # regex replace consumed "multistringyaml" wholesale
+        assert "multistringyaml" not in value.value
+        assert "configuration1" in value.value
+
+    def test_generated_go_concatenates(self):
+        from operator_builder_amd.codegen import generate_node
+
+        docs, _ = inspect_for_yaml(self.SRC, MarkerType.FIELD)
+        code = generate_node(docs[0], "resourceObj")
+        assert "parent.Spec.ConfigOption" in code
+        assert "parent.Spec.YamlType" in code
+        assert '" + parent.Spec.ConfigOption + "' in code
