@@ -365,6 +365,14 @@ def build_parser() -> argparse.ArgumentParser:
 
 def main(argv=None) -> int:
     parser = build_parser()
+
+    if argv is None:
+        argv = sys.argv[1:]
+    if not argv:
+        # bare invocation prints help (kubebuilder CLI behavior)
+        parser.print_help()
+        return 0
+
     args = parser.parse_args(argv)
     try:
         return args.func(args)
