@@ -194,3 +194,14 @@ def test_collection_create_api_idempotent(project):
 
     kind_file = read(project, "apis/apps/webapp.go")
     assert kind_file.count("v1alpha1apps.GroupVersion,") == 1
+
+
+def test_collection_itself_has_no_collection_ref(project):
+    # nested collections are unsupported: the collection's own spec gets
+    # no `collection` block (reference workload.go needsCollectionRef)
+    content = read(project, "apis/platforms/v1alpha1/cloudplatform_types.go")
+    assert "CollectionSpec" not in content
+    sample = read(
+        project, "config/samples/platforms_v1alpha1_cloudplatform.yaml"
+    )
+    assert "collection" not in sample
