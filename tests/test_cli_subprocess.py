@@ -135,3 +135,15 @@ spec:
     )
     assert result.returncode == 1
     assert "FATAL" in result.stderr
+
+
+def test_version_flag():
+    result = run_cli("--version")
+    assert result.returncode == 0
+    assert "0.1.0" in result.stdout
+
+
+def test_bare_invocation_prints_help():
+    result = run_cli()
+    assert result.returncode == 0
+    assert "usage: operator-builder" in result.stdout
