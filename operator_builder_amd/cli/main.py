@@ -16,6 +16,7 @@ import os
 import sys
 
 from .. import __version__
+from ..errors import OperatorBuilderError
 from ..license import (
     update_existing_source_header,
     update_project_license,
@@ -33,7 +34,7 @@ from ..workload import config as workload_config
 from ..workload import kinds, subcommand
 
 
-class CLIError(Exception):
+class CLIError(OperatorBuilderError):
     pass
 
 
@@ -376,13 +377,10 @@ def main(argv=None) -> int:
     args = parser.parse_args(argv)
     try:
         return args.func(args)
-    except (
-        CLIError,
-        ProjectError,
-        workload_config.ConfigError,
-        kinds.WorkloadConfigError,
-        subcommand.InitConfigError,
-    ) as err:
+    except OperatorBuilderError as err:
+        # every pipeline error type derives from OperatorBuilderError, so
+        # marker/manifest/RBAC/API-field/codegen failures all surface as
+        # the reference's single FATAL line (cmd/operator-builder/main.go:13-22)
         print(f"FATAL: {err}", file=sys.stderr)
         return 1
 

@@ -35,8 +35,10 @@ from ..yamlast.node import (
     TAG_VAR,
 )
 
+from ..errors import OperatorBuilderError
 
-class GenerateError(Exception):
+
+class GenerateError(OperatorBuilderError):
     pass
 
 

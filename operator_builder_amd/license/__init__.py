@@ -10,8 +10,10 @@ from __future__ import annotations
 import os
 import urllib.request
 
+from ..errors import OperatorBuilderError
 
-class LicenseError(Exception):
+
+class LicenseError(OperatorBuilderError):
     pass
 
 

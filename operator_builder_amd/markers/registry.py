@@ -13,8 +13,10 @@ from __future__ import annotations
 from dataclasses import dataclass, field
 from typing import Any, Callable, Optional
 
+from ..errors import OperatorBuilderError
 
-class MarkerError(Exception):
+
+class MarkerError(OperatorBuilderError):
     """A marker-language error (lexing, parsing, or inflation)."""
 
 

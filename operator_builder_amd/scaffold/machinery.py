@@ -18,8 +18,10 @@ import enum
 import os
 from dataclasses import dataclass, field
 
+from ..errors import OperatorBuilderError
 
-class ScaffoldError(Exception):
+
+class ScaffoldError(OperatorBuilderError):
     pass
 
 

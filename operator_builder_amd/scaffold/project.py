@@ -18,6 +18,8 @@ import yaml
 from ..workload.config import PLUGIN_KEY, PluginConfig
 from ..workload.kinds import Resource
 
+from ..errors import OperatorBuilderError
+
 PROJECT_FILE = "PROJECT"
 PROJECT_VERSION = "3"
 
@@ -28,7 +30,7 @@ LAYOUT = [
 ]
 
 
-class ProjectError(Exception):
+class ProjectError(OperatorBuilderError):
     pass
 
 

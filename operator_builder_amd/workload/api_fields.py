@@ -15,8 +15,10 @@ from typing import Any, Optional
 from ..utils import go_title
 from .markers import FieldType
 
+from ..errors import OperatorBuilderError
 
-class APIFieldError(Exception):
+
+class APIFieldError(OperatorBuilderError):
     pass
 
 

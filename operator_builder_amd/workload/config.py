@@ -15,8 +15,10 @@ import yaml
 from ..utils import glob as util_glob
 from . import kinds
 
+from ..errors import OperatorBuilderError
 
-class ConfigError(Exception):
+
+class ConfigError(OperatorBuilderError):
     pass
 
 

@@ -21,6 +21,8 @@ from ..utils import to_package_name, regular_plural
 from ..yamlast import emit_document, to_plain
 from ..yamlast.node import SCALAR, TAG_VAR
 
+from ..errors import OperatorBuilderError
+
 
 def _rewrite_collection_refs(doc) -> None:
     """Convert collection-marker substitutions to parent substitutions on
@@ -56,7 +58,7 @@ SAMPLE_WORKLOAD_API_KIND = "MyApp"
 SAMPLE_WORKLOAD_API_VERSION = "v1alpha1"
 
 
-class WorkloadConfigError(Exception):
+class WorkloadConfigError(OperatorBuilderError):
     pass
 
 

@@ -27,8 +27,10 @@ from .markers import (
     inspect_for_yaml,
 )
 
+from ..errors import OperatorBuilderError
 
-class ManifestError(Exception):
+
+class ManifestError(OperatorBuilderError):
     pass
 
 

@@ -16,6 +16,8 @@ from typing import Any
 
 from ..utils import regular_plural
 
+from ..errors import OperatorBuilderError
+
 CORE_GROUP = "core"
 KUBEBUILDER_PREFIX = "// +kubebuilder:rbac"
 
@@ -34,7 +36,7 @@ DEFAULT_STATUS_VERBS = ["get", "update", "patch"]
 _KNOWN_IRREGULARS = {"resourcequota": "resourcequotas"}
 
 
-class RBACError(Exception):
+class RBACError(OperatorBuilderError):
     pass
 
 

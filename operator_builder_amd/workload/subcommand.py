@@ -18,6 +18,8 @@ from .companion import CLI
 from .config import Processor
 from .markers import MarkerCollection
 
+from ..errors import OperatorBuilderError
+
 
 def init(processor: Processor) -> None:
     processor.workload.set_names()
@@ -90,7 +92,7 @@ class _IndentedDumper(yaml.SafeDumper):
         return super().increase_indent(flow, False)
 
 
-class InitConfigError(Exception):
+class InitConfigError(OperatorBuilderError):
     pass
 
 

@@ -6,12 +6,21 @@ driver invokes it, and that exactly one JSON line is emitted by rank 0.
 
 import json
 import os
+import socket
 import subprocess
 import sys
 
 import pytest
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _free_port() -> int:
+    # ask the kernel for an ephemeral port so the test can't collide with
+    # another process holding a fixed rendezvous port
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
 
 
 def run_bench(nproc: int, steps: int = 2, warmup: int = 0) -> dict:
@@ -25,7 +34,7 @@ def run_bench(nproc: int, steps: int = 2, warmup: int = 0) -> dict:
             "--master-addr",
             "127.0.0.1",
             "--master-port",
-            "29517",
+            str(_free_port()),
             os.path.join(REPO, "bench.py"),
         ]
     else:

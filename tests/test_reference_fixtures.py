@@ -99,25 +99,23 @@ def test_collection_case(tmp_path):
     workdir = generate(tmp_path, "collection", "github.com/acme/platform")
 
     # the collection and all three components scaffold
+    # (component GVKs from the reference fixture's *-component.yaml files:
+    # tenancy/TenancyCommon, tenancy/NsOperator, ingress/Contour)
     for path in [
         "apis/platforms/v1alpha1/cloudnativeplatform_types.go",
         "controllers/platforms/cloudnativeplatform_controller.go",
-        "apis/tenancy/v1alpha1/tenancynamespaceoperator_types.go"
-        if os.path.exists(
-            os.path.join(
-                workdir,
-                "apis/tenancy/v1alpha1/tenancynamespaceoperator_types.go",
-            )
-        )
-        else None,
+        "apis/tenancy/v1alpha1/tenancycommon_types.go",
+        "apis/tenancy/v1alpha1/nsoperator_types.go",
+        "apis/ingress/v1alpha1/contour_types.go",
+        "controllers/tenancy/tenancycommon_controller.go",
+        "controllers/tenancy/nsoperator_controller.go",
+        "controllers/ingress/contour_controller.go",
     ]:
-        if path:
-            assert os.path.exists(os.path.join(workdir, path)), path
+        assert os.path.exists(os.path.join(workdir, path)), path
 
-    # every component produced a controller dir entry
-    controllers = os.listdir(os.path.join(workdir, "controllers"))
-    assert "platforms" in controllers
-    assert len(controllers) >= 2
+    # every component group produced a controller dir entry
+    controllers = set(os.listdir(os.path.join(workdir, "controllers")))
+    assert {"platforms", "tenancy", "ingress"} <= controllers
 
 
 def test_edge_collection_case(tmp_path):
