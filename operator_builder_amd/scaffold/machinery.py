@@ -119,20 +119,45 @@ class Scaffold:
             f.write(content)
 
 
+def _normalized_lines(text: str) -> list[str]:
+    return [line.strip() for line in text.rstrip("\n").split("\n")]
+
+
+def _fragment_present(frag: str, file_norm: list[str]) -> bool:
+    """True iff the fragment's whitespace-normalized line sequence appears
+    as a contiguous run of lines in the file.
+
+    This is kubebuilder machinery's filterExistingValues semantics
+    (trimmed line-for-line equality) extended to multi-line fragments:
+    a fragment whose text merely appears as a *substring* of some longer
+    unrelated line (e.g. an import path that is a prefix of another)
+    must still be inserted.
+    """
+    needle = _normalized_lines(frag)
+    if not any(needle):
+        return False
+    n = len(needle)
+    for i in range(len(file_norm) - n + 1):
+        if file_norm[i : i + n] == needle:
+            return True
+    return False
+
+
 def insert_code_fragments(
     content: str, fragments: dict[Marker, list[str]]
 ) -> str:
     """Insert each fragment immediately before its marker line, skipping
-    fragments already present anywhere in the file (kubebuilder's
-    dedupe-on-insert semantics)."""
+    fragments already present in the file (kubebuilder's dedupe-on-insert
+    semantics: whitespace-normalized line comparison, not substring)."""
     lines = content.split("\n")
 
     for marker, frags in fragments.items():
         marker_text = str(marker)
+        file_norm = _normalized_lines(content)
 
         to_insert = []
         for frag in frags:
-            if frag.rstrip("\n") and frag.rstrip("\n") in content:
+            if _fragment_present(frag, file_norm):
                 continue
             if frag not in to_insert:
                 to_insert.append(frag)
