@@ -118,14 +118,41 @@ def cmd_create_api(args) -> int:
 
     workload = processor.workload
 
+    ctx = _build_context(base_dir, project, workload)
+
+    # kubebuilder semantics: re-scaffolding an API whose GVK is already
+    # recorded in the PROJECT file requires --force
+    # (docs/api-updates-upgrades.md:20-36 documents the
+    # `--controller=false --resource --force` update workflow)
+    if not args.force:
+        for w in processor.get_workloads():
+            existing = project.get_resource(
+                w.get_api_group(), w.get_api_version(), w.get_api_kind()
+            )
+            if existing is not None:
+                raise CLIError(
+                    "failed to create API: API resource already exists "
+                    f"for {w.get_api_group()}/{w.get_api_version()} "
+                    f"{w.get_api_kind()}; re-run with --force to "
+                    "regenerate it"
+                )
+
     print("Building API...")
 
-    ctx = _build_context(base_dir, project, workload)
-    scaffold_api(base_dir, ctx, workload)
+    scaffold_api(
+        base_dir,
+        ctx,
+        workload,
+        controller=args.controller,
+        resource=args.resource,
+    )
 
     # record every scaffolded resource in the PROJECT file
     for w in processor.get_workloads():
-        project.add_resource(resource_for_workload(ctx, w))
+        res = resource_for_workload(ctx, w)
+        res.controller = args.controller
+        res.has_api = args.resource
+        project.add_resource(res)
     project.plugin_config.workload_config_path = config_path
     project.save(base_dir)
 

@@ -78,6 +78,7 @@ class Project:
                     controller=bool(res.get("controller", False)),
                     crd_version=api.get("crdVersion", "v1"),
                     namespaced=bool(api.get("namespaced", False)),
+                    has_api="api" in res,
                 )
             )
 
@@ -93,21 +94,25 @@ class Project:
             "repo": self.repo,
         }
         if self.resources:
-            raw["resources"] = [
-                {
-                    "api": {
+            raw["resources"] = []
+            for res in self.resources:
+                entry: dict = {}
+                if res.has_api:
+                    entry["api"] = {
                         "crdVersion": res.crd_version,
                         "namespaced": res.namespaced,
-                    },
-                    "controller": res.controller,
-                    "domain": res.domain,
-                    "group": res.group,
-                    "kind": res.kind,
-                    "path": res.path,
-                    "version": res.version,
-                }
-                for res in self.resources
-            ]
+                    }
+                entry.update(
+                    {
+                        "controller": res.controller,
+                        "domain": res.domain,
+                        "group": res.group,
+                        "kind": res.kind,
+                        "path": res.path,
+                        "version": res.version,
+                    }
+                )
+                raw["resources"].append(entry)
         raw["version"] = PROJECT_VERSION
 
         with open(

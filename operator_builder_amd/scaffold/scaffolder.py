@@ -68,14 +68,39 @@ def scaffold_init(base_dir: str, ctx: Context, workload: Workload) -> None:
     )
 
 
-def scaffold_api(base_dir: str, ctx: Context, workload: Workload) -> None:
+def scaffold_api(
+    base_dir: str,
+    ctx: Context,
+    workload: Workload,
+    *,
+    controller: bool = True,
+    resource: bool = True,
+) -> None:
     """API + controller generation for a workload tree
-    (reference scaffolds/api.go:84-193)."""
-    _scaffold_workload(Scaffold(base_dir), ctx, workload)
+    (reference scaffolds/api.go:84-193).
+
+    ``controller``/``resource`` implement the kubebuilder-level
+    `create api --controller/--resource` gating the reference inherits
+    from its plugin bundle (docs/api-updates-upgrades.md:20-36):
+    `--controller=false` skips generating controller code while
+    `--resource` regenerates the API.
+    """
+    _scaffold_workload(
+        Scaffold(base_dir),
+        ctx,
+        workload,
+        controller=controller,
+        resource=resource,
+    )
 
 
 def _scaffold_workload(
-    scaffold: Scaffold, ctx: Context, workload: Workload
+    scaffold: Scaffold,
+    ctx: Context,
+    workload: Workload,
+    *,
+    controller: bool = True,
+    resource: bool = True,
 ) -> None:
     # components swap in their own resource so Group/Version/Kind come
     # from the child, not the parent (reference api.go:117-127)
@@ -83,48 +108,63 @@ def _scaffold_workload(
         resource_for_workload(ctx, workload), workload
     )
 
-    _scaffold_api_dir(scaffold, workload_ctx, workload)
+    if resource:
+        _scaffold_api_dir(scaffold, workload_ctx, workload)
 
-    # controller + user-editable stubs + crd kustomization entry
-    scaffold.execute(
-        controller_tpl.controller(workload_ctx, workload),
-        controller_tpl.phases(workload_ctx),
-        internal_tpl.dependencies_component(workload_ctx),
-        internal_tpl.mutate_component(workload_ctx),
-        internal_tpl.crd_kustomization(workload_ctx),
-    )
+    if controller:
+        # controller + user-editable stubs + crd kustomization entry
+        scaffold.execute(
+            controller_tpl.controller(workload_ctx, workload),
+            controller_tpl.phases(workload_ctx),
+            internal_tpl.dependencies_component(workload_ctx),
+            internal_tpl.mutate_component(workload_ctx),
+            internal_tpl.crd_kustomization(workload_ctx),
+        )
 
-    # suite test for the controller group (upstream golang/v3 behavior)
-    suite_file, suite_fragments = controller_tpl.suite_test(workload_ctx)
-    scaffold.execute(suite_file, suite_fragments)
+        # suite test for the controller group (upstream golang/v3 behavior)
+        suite_file, suite_fragments = controller_tpl.suite_test(workload_ctx)
+        scaffold.execute(suite_file, suite_fragments)
 
     # wire the new api + controller into main.go
-    scaffold.execute(root_tpl.main_updater(workload_ctx))
-
-    # crd sample + per-kind rbac roles
     scaffold.execute(
-        internal_tpl.crd_sample(
+        root_tpl.main_updater(
             workload_ctx,
-            workload.get_api_spec_fields(),
-            workload.is_cluster_scoped(),
-        ),
-        base_tpl.rbac_editor_role(workload_ctx),
-        base_tpl.rbac_viewer_role(workload_ctx),
+            wire_resource=resource,
+            wire_controller=controller,
+        )
     )
 
-    # e2e workload test
-    scaffold.execute(e2e_tpl.workload_test(workload_ctx, workload))
+    if resource:
+        # crd sample + per-kind rbac roles
+        scaffold.execute(
+            internal_tpl.crd_sample(
+                workload_ctx,
+                workload.get_api_spec_fields(),
+                workload.is_cluster_scoped(),
+            ),
+            base_tpl.rbac_editor_role(workload_ctx),
+            base_tpl.rbac_viewer_role(workload_ctx),
+        )
 
-    # companion CLI subcommands — only for workloads that belong to a
-    # companion CLI (a later standalone added to a project that has one
-    # may itself define none)
-    if ctx.cli_root_command_name and workload.get_root_command().name:
-        _scaffold_cli(scaffold, workload_ctx, workload)
+        # e2e workload test
+        scaffold.execute(e2e_tpl.workload_test(workload_ctx, workload))
+
+        # companion CLI subcommands — only for workloads that belong to a
+        # companion CLI (a later standalone added to a project that has one
+        # may itself define none)
+        if ctx.cli_root_command_name and workload.get_root_command().name:
+            _scaffold_cli(scaffold, workload_ctx, workload)
 
     # recurse into collection components
     if workload.is_collection():
         for component in workload.get_components():
-            _scaffold_workload(scaffold, ctx, component)
+            _scaffold_workload(
+                scaffold,
+                ctx,
+                component,
+                controller=controller,
+                resource=resource,
+            )
 
 
 def _scaffold_api_dir(

@@ -151,35 +151,50 @@ func main() {{
     return File("main.go", content, IfExists.OVERWRITE)
 
 
-def main_updater(ctx: Context) -> Fragments:
+def main_updater(
+    ctx: Context,
+    *,
+    wire_resource: bool = True,
+    wire_controller: bool = True,
+) -> Fragments:
     """Wire a new API + controller into main.go at the scaffold markers
-    (reference MainUpdater, templates/main.go:44-160)."""
+    (reference MainUpdater, templates/main.go:44-160; the
+    WireResource/WireController knobs mirror machinery's MainUpdater
+    fields set at scaffolds/api.go:154-159)."""
     res = ctx.resource
 
-    imports = [
-        f'{res.import_alias} "{res.path}"\n',
-        f'{res.group}controllers "{ctx.repo}/controllers/{res.group}"\n'
-        if ctx.multi_group and res.group
-        else f'"{ctx.repo}/controllers"\n',
-    ]
+    imports: list[str] = []
+    add_scheme: list[str] = []
+    setup: list[str] = []
 
-    add_scheme = [
-        f"utilruntime.Must({res.import_alias}.AddToScheme(scheme))\n"
-    ]
+    if wire_resource:
+        imports.append(f'{res.import_alias} "{res.path}"\n')
+        add_scheme.append(
+            f"utilruntime.Must({res.import_alias}.AddToScheme(scheme))\n"
+        )
 
-    if ctx.multi_group and res.group:
-        setup = [f"{res.group}controllers.New{res.kind}Reconciler(mgr),\n"]
-    else:
-        setup = [f"controllers.New{res.kind}Reconciler(mgr),\n"]
+    if wire_controller:
+        if ctx.multi_group and res.group:
+            imports.append(
+                f'{res.group}controllers '
+                f'"{ctx.repo}/controllers/{res.group}"\n'
+            )
+            setup.append(
+                f"{res.group}controllers.New{res.kind}Reconciler(mgr),\n"
+            )
+        else:
+            imports.append(f'"{ctx.repo}/controllers"\n')
+            setup.append(f"controllers.New{res.kind}Reconciler(mgr),\n")
 
-    return Fragments(
-        path="main.go",
-        fragments={
-            IMPORT_MARKER: imports,
-            SCHEME_MARKER: add_scheme,
-            RECONCILER_MARKER: setup,
-        },
-    )
+    fragments: dict = {}
+    if imports:
+        fragments[IMPORT_MARKER] = imports
+    if add_scheme:
+        fragments[SCHEME_MARKER] = add_scheme
+    if setup:
+        fragments[RECONCILER_MARKER] = setup
+
+    return Fragments(path="main.go", fragments=fragments)
 
 
 def go_mod(ctx: Context) -> File:

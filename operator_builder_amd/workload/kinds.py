@@ -370,6 +370,9 @@ class Resource:
     crd_version: str = "v1"
     namespaced: bool = True
     controller: bool = True
+    # False when `create api --resource=false` recorded a controller-only
+    # entry (kubebuilder PROJECT files omit the api: block in that case)
+    has_api: bool = True
 
     @property
     def import_alias(self) -> str:

@@ -94,3 +94,88 @@ def test_version_bump_extends_registries(project):
     proj = read(project, "PROJECT")
     assert "version: v1alpha1" in proj
     assert "version: v1alpha2" in proj
+
+
+def test_create_api_without_force_conflicts(project, capsys):
+    # the project fixture already scaffolded apps/v1alpha1 BookStore:
+    # re-running without --force must error, not silently overwrite
+    # (kubebuilder golang/v3 InjectResource check)
+    assert main(["create", "api"]) == 1
+    err = capsys.readouterr().err
+    assert "FATAL" in err
+    assert "already exists" in err
+
+
+def test_update_workflow_controller_false_resource_force(project):
+    """The documented update workflow: `create api --controller=false
+    --resource --force` regenerates the API without touching controller
+    code (reference docs/api-updates-upgrades.md:20-36)."""
+    controller_path = "controllers/apps/bookstore_controller.go"
+    before = read(project, controller_path)
+
+    # make the controller file distinguishable so an overwrite is visible
+    full = os.path.join(project, controller_path)
+    with open(full, "w", encoding="utf-8") as f:
+        f.write("// user-modified controller\n" + before)
+
+    # evolve the manifest so the regenerated API visibly changes
+    cfg_dir = os.path.join(project, ".workloadConfig")
+    res_file = None
+    for name in os.listdir(cfg_dir):
+        if name != "workload.yaml" and name.endswith(".yaml"):
+            res_file = os.path.join(cfg_dir, name)
+            break
+    assert res_file
+
+    assert (
+        main(["create", "api", "--controller=false", "--resource", "--force"])
+        == 0
+    )
+
+    # API regenerated
+    assert os.path.exists(
+        os.path.join(project, "apis/apps/v1alpha1/bookstore_types.go")
+    )
+    # controller untouched (user modification preserved)
+    after = read(project, controller_path)
+    assert after.startswith("// user-modified controller\n")
+
+    # PROJECT records controller: false for the updated resource
+    proj = read(project, "PROJECT")
+    assert "controller: false" in proj
+
+
+def test_create_api_resource_false(tmp_path, monkeypatch):
+    """--resource=false skips API templates but still scaffolds the
+    controller (kubebuilder --resource semantics)."""
+    workdir = tmp_path / "ctrl-only"
+    workdir.mkdir()
+    shutil.copytree(
+        os.path.join(FIXTURES, "standalone"), workdir / ".workloadConfig"
+    )
+    monkeypatch.chdir(workdir)
+    assert (
+        main(
+            [
+                "init",
+                "--workload-config",
+                ".workloadConfig/workload.yaml",
+                "--repo",
+                "github.com/acme/bookstore",
+            ]
+        )
+        == 0
+    )
+    assert main(["create", "api", "--resource=false"]) == 0
+
+    assert not os.path.exists(
+        str(workdir / "apis/apps/v1alpha1/bookstore_types.go")
+    )
+    assert os.path.exists(
+        str(workdir / "controllers/apps/bookstore_controller.go")
+    )
+
+    # PROJECT entry has no api: block but records the controller
+    proj = read(str(workdir), "PROJECT")
+    assert "controller: true" in proj
+    assert "crdVersion" not in proj

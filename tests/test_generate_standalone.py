@@ -177,8 +177,9 @@ def test_sample_manifest(project):
 
 
 def test_create_api_idempotent(project):
-    # a second run must not duplicate inserted fragments
-    assert main(["create", "api"]) == 0
+    # a second run must not duplicate inserted fragments (--force is
+    # required to regenerate an existing API, kubebuilder semantics)
+    assert main(["create", "api", "--force"]) == 0
 
     content = read(project, "main.go")
     assert (

@@ -73,7 +73,7 @@ def test_license_flow(tmp_path, monkeypatch):
         assert f.read().startswith("// Copyright NewCo 2026.")
 
     # regeneration picks up the new boilerplate
-    assert main(["create", "api"]) == 0
+    assert main(["create", "api", "--force"]) == 0
     with open("apis/apps/v1alpha1/bookstore_types.go") as f:
         assert f.read().startswith("// Copyright NewCo 2026.")
 

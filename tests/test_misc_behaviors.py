@@ -183,7 +183,7 @@ spec:
         '  a: "1"  # +operator-builder:field:name=alpha,type=string\n'
         '  b: "2"  # +operator-builder:field:name=beta,type=string,default="2"\n',
     )
-    assert main(["create", "api"]) == 0
+    assert main(["create", "api", "--force"]) == 0
 
     types = read(root, "apis/apps/v1/app_types.go")
     assert "Alpha string" in types
