@@ -1,0 +1,233 @@
+"""goimports-lite: the subset of ``imports.Process`` that changes the
+reference's generated output.
+
+kubebuilder machinery formats every ``.go`` file it scaffolds with
+``golang.org/x/tools/imports`` (goimports), which — on the
+mostly-formatted text the templates render — has three visible effects:
+
+  1. unused imports are removed (this is why the reference template's
+     unused ``sigs.k8s.io/controller-runtime/pkg/controller`` import in
+     templates/main.go:178 never appears in real output);
+  2. imports inside each blank-line-separated group are sorted by path;
+  3. gofmt hygiene: no trailing whitespace, at most one consecutive
+     blank line, exactly one trailing newline.
+
+This module reproduces those effects.  Full gofmt (re-indentation,
+alignment) is NOT attempted: the templates render already-gofmt-shaped
+text, and PARITY.md documents the residual risk.
+"""
+
+from __future__ import annotations
+
+import re
+
+from .lexer import Token, tokenize, GoLexError
+
+# import spec inside a block: [alias] "path" [// comment]
+_IMPORT_LINE = re.compile(
+    r'^\s*(?P<alias>[\w.]+\s+)?"(?P<path>[^"]+)"\s*(?P<comment>//.*)?$'
+)
+
+_VERSION_ELEM = re.compile(r"^v\d+$")
+
+
+def _default_name(path: str) -> str | None:
+    """The identifier an unaliased import binds, inferred from the path.
+
+    Returns None when inference is unsafe (the last element is a version
+    like ``v3``), in which case the import is never treated as unused.
+    """
+    last = path.rstrip("/").rsplit("/", 1)[-1]
+    # gopkg.in/yaml.v2 -> yaml
+    if "." in last:
+        base, _, suffix = last.rpartition(".")
+        if _VERSION_ELEM.match(suffix) and base:
+            return base
+    if _VERSION_ELEM.match(last):
+        return None
+    # go-playground/validator -> validator (dashes are not identifiers;
+    # package names conventionally drop them — unsafe to infer)
+    if not last.isidentifier():
+        return None
+    return last
+
+
+def _used_identifiers(tokens: list[Token]) -> set[str]:
+    """Identifiers that appear in qualified-selector position
+    (``name.Something``) or as a bare identifier anywhere outside the
+    import declaration — conservative: shadowing keeps an import."""
+    used: set[str] = set()
+    sig = [t for t in tokens if t.kind not in ("NEWLINE", "COMMENT")]
+
+    # find the span of import declarations to exclude them from usage
+    skip: set[int] = set()
+    i = 0
+    while i < len(sig):
+        t = sig[i]
+        if t.kind == "KEYWORD" and t.text == "import":
+            skip.add(i)
+            j = i + 1
+            if j < len(sig) and sig[j].text == "(":
+                depth = 0
+                while j < len(sig):
+                    skip.add(j)
+                    if sig[j].text == "(":
+                        depth += 1
+                    elif sig[j].text == ")":
+                        depth -= 1
+                        if depth == 0:
+                            break
+                    j += 1
+                i = j + 1
+                continue
+            # single import: alias? "path"
+            while j < len(sig) and sig[j].kind in ("IDENT", "STRING", "OP"):
+                skip.add(j)
+                if sig[j].kind == "STRING":
+                    break
+                j += 1
+            i = j + 1
+            continue
+        i += 1
+
+    for idx, t in enumerate(sig):
+        if idx in skip or t.kind != "IDENT":
+            continue
+        # selector member position (`x.THIS`) is not a package use
+        if idx > 0 and sig[idx - 1].kind == "OP" and sig[idx - 1].text == ".":
+            continue
+        used.add(t.text)
+    return used
+
+
+def _raw_string_lines(tokens: list[Token]) -> set[int]:
+    lines: set[int] = set()
+    for t in tokens:
+        if t.kind == "RAW_STRING":
+            span = t.text.count("\n")
+            # interior lines (and the first/last) must not be rewritten
+            for ln in range(t.line, t.line + span + 1):
+                lines.add(ln)
+    return lines
+
+
+def format_go(src: str) -> str:
+    """Apply the goimports-lite pipeline; returns src unchanged if the
+    file does not tokenize (never corrupt output on a lexer gap)."""
+    try:
+        tokens = tokenize(src)
+    except GoLexError:
+        return src
+
+    used = _used_identifiers(tokens)
+    protected = _raw_string_lines(tokens)
+
+    lines = src.split("\n")
+    out: list[str] = []
+    i = 0
+    n = len(lines)
+
+    while i < n:
+        line = lines[i]
+        lineno = i + 1
+        stripped = line.strip()
+
+        if stripped.startswith("import (") and lineno not in protected:
+            # collect the block
+            block: list[str] = []
+            i += 1
+            while i < n and lines[i].strip() != ")":
+                block.append(lines[i])
+                i += 1
+            closing = lines[i] if i < n else ")"
+            i += 1
+
+            out.append(line)
+            out.extend(_rewrite_import_block(block, used))
+            out.append(closing)
+            continue
+
+        out.append(line)
+        i += 1
+
+    # hygiene: trailing whitespace + blank-line collapse (raw-string
+    # interiors excluded), exactly one trailing newline
+    cleaned: list[str] = []
+    blank_run = 0
+    for idx, line in enumerate(out):
+        lineno = idx + 1
+        if lineno in protected:
+            cleaned.append(line)
+            blank_run = 0
+            continue
+        line = line.rstrip()
+        if line == "":
+            blank_run += 1
+            if blank_run > 1:
+                continue
+        else:
+            blank_run = 0
+        cleaned.append(line)
+
+    text = "\n".join(cleaned)
+    return text.rstrip("\n") + "\n"
+
+
+def _rewrite_import_block(block: list[str], used: set[str]) -> list[str]:
+    """Drop unused specs and sort each group by path.
+
+    Blank lines AND comment lines (e.g. ``//+operator-builder:...``
+    scaffold markers) are group barriers: sorting never moves an import
+    across them, and a comment line keeps its exact position.
+    """
+    # items: ("group", [(path, line), ...]) | ("comment", line) | ("blank",)
+    items: list[tuple] = []
+
+    def group() -> list[tuple[str, str]]:
+        if not items or items[-1][0] != "group":
+            items.append(("group", []))
+        return items[-1][1]
+
+    for line in block:
+        if line.strip() == "":
+            items.append(("blank",))
+            continue
+        m = _IMPORT_LINE.match(line)
+        if not m:
+            items.append(("comment", "\t" + line.strip()))
+            continue
+        alias = (m.group("alias") or "").strip()
+        path = m.group("path")
+
+        if alias in ("_", "."):
+            name: str | None = alias  # always kept
+        elif alias:
+            name = alias
+        else:
+            name = _default_name(path)
+
+        if name not in ("_", ".") and name is not None and name not in used:
+            continue  # unused -> removed (goimports behavior)
+
+        group().append((path, "\t" + line.strip()))
+
+    result: list[str] = []
+    pending_blank = False
+    emitted = False
+    for item in items:
+        if item[0] == "blank":
+            pending_blank = True
+            continue
+        if item[0] == "group" and not item[1]:
+            continue  # fully-removed group: its separator goes with it
+        if pending_blank and emitted:
+            result.append("")
+        pending_blank = False
+        if item[0] == "comment":
+            result.append(item[1])
+        else:
+            result.extend(
+                line for _, line in sorted(item[1], key=lambda p: p[0])
+            )
+        emitted = True
+    return result

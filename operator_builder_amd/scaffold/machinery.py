@@ -18,6 +18,8 @@ import enum
 import os
 from dataclasses import dataclass, field
 
+from ..golang import format_go
+
 from ..errors import OperatorBuilderError
 
 
@@ -95,8 +97,15 @@ class Scaffold:
                 )
 
         self._ensure_dir(full)
+        content = item.content
+        if item.path.endswith(".go"):
+            # kubebuilder machinery formats every scaffolded .go file
+            # with goimports (imports.Process); format_go reproduces the
+            # output-visible subset (unused-import removal, in-group
+            # sorting, gofmt hygiene)
+            content = format_go(content)
         with open(full, "w", encoding="utf-8") as f:
-            f.write(item.content)
+            f.write(content)
 
     def _insert_fragments(self, item: Fragments) -> None:
         full = self._full(item.path)
@@ -114,6 +123,8 @@ class Scaffold:
             content = f.read()
 
         content = insert_code_fragments(content, item.fragments)
+        if item.path.endswith(".go"):
+            content = format_go(content)
 
         with open(full, "w", encoding="utf-8") as f:
             f.write(content)

@@ -19,7 +19,7 @@ package apps
 import (
 	v1alpha1apps "github.com/acme/bookstore/apis/apps/v1alpha1"
 	//+operator-builder:imports
-	
+
 	"k8s.io/apimachinery/pkg/runtime/schema"
 )
 

@@ -19,8 +19,8 @@ package bookstore
 import (
 	"fmt"
 
-	"sigs.k8s.io/yaml"
 	"sigs.k8s.io/controller-runtime/pkg/client"
+	"sigs.k8s.io/yaml"
 
 	"github.com/nukleros/operator-builder-tools/pkg/controller/workload"
 

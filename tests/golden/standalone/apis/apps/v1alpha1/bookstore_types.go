@@ -19,8 +19,8 @@ package v1alpha1
 import (
 	"errors"
 
-	"github.com/nukleros/operator-builder-tools/pkg/status"
 	"github.com/nukleros/operator-builder-tools/pkg/controller/workload"
+	"github.com/nukleros/operator-builder-tools/pkg/status"
 	metav1 "k8s.io/apimachinery/pkg/apis/meta/v1"
 	"k8s.io/apimachinery/pkg/runtime/schema"
 )
@@ -29,7 +29,6 @@ var ErrUnableToConvertBookStore = errors.New("unable to convert to BookStore")
 
 // EDIT THIS FILE!  THIS IS SCAFFOLDING FOR YOU TO OWN!
 // NOTE: json tags are required.  Any new fields you add must have json tags for the fields to be serialized.
-
 
 // BookStoreSpec defines the desired state of BookStore.
 type BookStoreSpec struct {
@@ -104,7 +103,6 @@ Name string `json:"name,omitempty"`
 TargetPort int `json:"targetPort,omitempty"`
 
 }
-
 
 // BookStoreStatus defines the observed state of BookStore.
 type BookStoreStatus struct {

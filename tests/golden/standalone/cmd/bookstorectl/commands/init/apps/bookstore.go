@@ -24,8 +24,8 @@ import (
 
 	"github.com/acme/bookstore/apis/apps"
 
-	cmdinit "github.com/acme/bookstore/cmd/bookstorectl/commands/init"
 	v1alpha1bookstore "github.com/acme/bookstore/apis/apps/v1alpha1/bookstore"
+	cmdinit "github.com/acme/bookstore/cmd/bookstorectl/commands/init"
 	//+operator-builder:imports
 )
 

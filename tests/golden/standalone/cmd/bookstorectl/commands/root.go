@@ -20,13 +20,13 @@ import (
 	"github.com/spf13/cobra"
 
 	// common imports for subcommands
-	cmdinit "github.com/acme/bookstore/cmd/bookstorectl/commands/init"
 	cmdgenerate "github.com/acme/bookstore/cmd/bookstorectl/commands/generate"
+	cmdinit "github.com/acme/bookstore/cmd/bookstorectl/commands/init"
 	cmdversion "github.com/acme/bookstore/cmd/bookstorectl/commands/version"
 
 	// specific imports for workloads
-	initapps "github.com/acme/bookstore/cmd/bookstorectl/commands/init/apps"
 	generateapps "github.com/acme/bookstore/cmd/bookstorectl/commands/generate/apps"
+	initapps "github.com/acme/bookstore/cmd/bookstorectl/commands/init/apps"
 	versionapps "github.com/acme/bookstore/cmd/bookstorectl/commands/version/apps"
 	//+operator-builder:subcommands:imports
 )

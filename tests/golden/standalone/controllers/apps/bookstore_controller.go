@@ -21,19 +21,19 @@ import (
 	"fmt"
 
 	"github.com/go-logr/logr"
+	"github.com/nukleros/operator-builder-tools/pkg/controller/phases"
+	"github.com/nukleros/operator-builder-tools/pkg/controller/predicates"
+	"github.com/nukleros/operator-builder-tools/pkg/controller/workload"
 	apierrs "k8s.io/apimachinery/pkg/api/errors"
 	"k8s.io/client-go/tools/record"
 	ctrl "sigs.k8s.io/controller-runtime"
 	"sigs.k8s.io/controller-runtime/pkg/client"
 	"sigs.k8s.io/controller-runtime/pkg/controller"
-	"github.com/nukleros/operator-builder-tools/pkg/controller/phases"
-	"github.com/nukleros/operator-builder-tools/pkg/controller/predicates"
-	"github.com/nukleros/operator-builder-tools/pkg/controller/workload"
 
-	"github.com/acme/bookstore/internal/dependencies"
-	"github.com/acme/bookstore/internal/mutate"
 	appsv1alpha1 "github.com/acme/bookstore/apis/apps/v1alpha1"
 	"github.com/acme/bookstore/apis/apps/v1alpha1/bookstore"
+	"github.com/acme/bookstore/internal/dependencies"
+	"github.com/acme/bookstore/internal/mutate"
 )
 
 // BookStoreReconciler reconciles a BookStore object.

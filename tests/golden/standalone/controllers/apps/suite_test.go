@@ -17,18 +17,18 @@ limitations under the License.
 package apps
 
 import (
-	"path/filepath"
-	"testing"
+	appsv1alpha1 "github.com/acme/bookstore/apis/apps/v1alpha1"
 	. "github.com/onsi/ginkgo"
 	. "github.com/onsi/gomega"
 	"k8s.io/client-go/kubernetes/scheme"
 	"k8s.io/client-go/rest"
+	"path/filepath"
 	"sigs.k8s.io/controller-runtime/pkg/client"
 	"sigs.k8s.io/controller-runtime/pkg/envtest"
 	"sigs.k8s.io/controller-runtime/pkg/envtest/printer"
 	logf "sigs.k8s.io/controller-runtime/pkg/log"
 	"sigs.k8s.io/controller-runtime/pkg/log/zap"
-	appsv1alpha1 "github.com/acme/bookstore/apis/apps/v1alpha1"
+	"testing"
 	//+kubebuilder:scaffold:imports
 )
 
