@@ -237,9 +237,12 @@ def check_tree(root: str) -> list[CheckIssue]:
                 pkg_names[rel] = m.group(1)
 
             # top-level plain functions (methods have a receiver and
-            # don't match this pattern); duplicates break compilation
+            # don't match this pattern); duplicates break compilation —
+            # except `init`, which Go allows any number of times
             for fm in func_decl.finditer(src):
                 name = fm.group(1)
+                if name == "init":
+                    continue
                 if name in funcs:
                     lineno = src[: fm.start()].count("\n") + 1
                     issues.append(
