@@ -20,6 +20,7 @@ text, and PARITY.md documents the residual risk.
 from __future__ import annotations
 
 import re
+from functools import lru_cache
 
 from .lexer import Token, tokenize, GoLexError
 
@@ -111,6 +112,87 @@ def _raw_string_lines(tokens: list[Token]) -> set[int]:
     return lines
 
 
+def _reindent(text: str) -> str:
+    """gofmt-style block re-indentation (tabs by bracket depth).
+
+    The raw template render loses/garbles indentation wherever control
+    actions owned the line ({{- .SourceCode }} splices, {{ end -}}
+    trims); gofmt recomputes indentation from syntax, so the oracle
+    must too.  Lines inside raw strings are untouched; `case`/`default`
+    dedent one level (gofmt switch style); continuation lines keep the
+    surrounding depth.
+    """
+    pass
+
+    try:
+        tokens = tokenize(text)
+    except GoLexError:
+        return text
+
+    protected: set[int] = set()
+    for t in tokens:
+        if t.kind in ("RAW_STRING", "COMMENT") and "\n" in t.text:
+            # interior + closing lines of raw strings and block comments
+            # keep their own indentation
+            for ln in range(t.line + 1, t.line + t.text.count("\n") + 1):
+                protected.add(ln)
+
+    by_line: dict[int, list] = {}
+    for t in tokens:
+        if t.kind == "OP" and t.text in "()[]{}":
+            by_line.setdefault(t.line, []).append(t.text)
+        elif t.kind not in ("NEWLINE",):
+            by_line.setdefault(t.line, []).append(None)
+
+    lines = text.split("\n")
+    out: list[str] = []
+    depth = 0
+    switch_depths: list[int] = []
+    for idx, raw in enumerate(lines):
+        lineno = idx + 1
+        s = raw.strip()
+        ops = by_line.get(lineno, [])
+
+        if lineno in protected or s == "":
+            out.append(raw if lineno in protected else "")
+            # still track depth from protected lines' brackets
+            for op in ops:
+                if op is None:
+                    continue
+                if op in "([{":
+                    depth += 1
+                else:
+                    depth = max(0, depth - 1)
+            continue
+
+        # leading run of closers dedents this line
+        leading_closers = 0
+        for op in ops:
+            if op is not None and op in ")]}":
+                leading_closers += 1
+            else:
+                break
+
+        indent = max(0, depth - leading_closers)
+        if s.startswith(("case ", "default:")) or s == "default:":
+            indent = max(0, indent - 1)
+        if s.startswith("switch ") or s.startswith("select {"):
+            switch_depths.append(depth)
+
+        out.append("\t" * indent + s)
+
+        for op in ops:
+            if op is None:
+                continue
+            if op in "([{":
+                depth += 1
+            else:
+                depth = max(0, depth - 1)
+
+    return "\n".join(out)
+
+
+@lru_cache(maxsize=512)
 def format_go(src: str) -> str:
     """Apply the goimports-lite pipeline; returns src unchanged if the
     file does not tokenize (never corrupt output on a lexer gap)."""
@@ -133,44 +215,38 @@ def format_go(src: str) -> str:
         stripped = line.strip()
 
         if stripped.startswith("import (") and lineno not in protected:
-            # collect the block
+            # collect the block; bail out verbatim if it doesn't look
+            # like a well-formed import block (never corrupt output)
             block: list[str] = []
-            i += 1
-            while i < n and lines[i].strip() != ")":
-                block.append(lines[i])
+            j = i + 1
+            well_formed = False
+            while j < n:
+                s = lines[j].strip()
+                if s == ")":
+                    well_formed = True
+                    break
+                if s and not s.startswith("//") and not _IMPORT_LINE.match(
+                    lines[j]
+                ):
+                    break
+                block.append(lines[j])
+                j += 1
+            if not well_formed:
+                out.append(line)
                 i += 1
-            closing = lines[i] if i < n else ")"
-            i += 1
+                continue
 
             out.append(line)
             out.extend(_rewrite_import_block(block, used))
-            out.append(closing)
+            out.append(lines[j])
+            i = j + 1
             continue
 
         out.append(line)
         i += 1
 
-    # hygiene: trailing whitespace + blank-line collapse (raw-string
-    # interiors excluded), exactly one trailing newline
-    cleaned: list[str] = []
-    blank_run = 0
-    for idx, line in enumerate(out):
-        lineno = idx + 1
-        if lineno in protected:
-            cleaned.append(line)
-            blank_run = 0
-            continue
-        line = line.rstrip()
-        if line == "":
-            blank_run += 1
-            if blank_run > 1:
-                continue
-        else:
-            blank_run = 0
-        cleaned.append(line)
-
-    text = "\n".join(cleaned)
-    return text.rstrip("\n") + "\n"
+    text = "\n".join(out)
+    return _final_pass(text, tokens if text == src else None)
 
 
 def _rewrite_import_block(block: list[str], used: set[str]) -> list[str]:
@@ -231,3 +307,88 @@ def _rewrite_import_block(block: list[str], used: set[str]) -> list[str]:
             )
         emitted = True
     return result
+
+
+def _final_pass(text: str, tokens: list[Token] | None = None) -> str:
+    """One-pass gofmt hygiene + re-indentation over one token stream:
+    trailing-whitespace strip, blank-line collapse (max one), gofmt's
+    drop-blank-before-closing-brace, and bracket-depth re-indentation.
+    Raw-string and block-comment interiors pass through verbatim."""
+    if tokens is None:
+        try:
+            tokens = tokenize(text)
+        except GoLexError:
+            return text
+
+    raw_protected: set[int] = set()
+    verbatim: set[int] = set()
+    for t in tokens:
+        if "\n" not in t.text:
+            continue
+        if t.kind == "RAW_STRING":
+            for ln in range(t.line, t.line + t.text.count("\n") + 1):
+                raw_protected.add(ln)
+                verbatim.add(ln)
+        elif t.kind == "COMMENT":
+            for ln in range(t.line + 1, t.line + t.text.count("\n") + 1):
+                verbatim.add(ln)
+
+    by_line: dict[int, list] = {}
+    for t in tokens:
+        if t.kind == "NEWLINE":
+            continue
+        if t.kind == "OP" and t.text in "()[]{}":
+            by_line.setdefault(t.line, []).append(t.text)
+        else:
+            by_line.setdefault(t.line, []).append(None)
+
+    def apply_depth(ops, depth):
+        for op in ops:
+            if op is None:
+                continue
+            if op in "([{":
+                depth += 1
+            else:
+                depth = max(0, depth - 1)
+        return depth
+
+    lines = text.split("\n")
+    out: list[str] = []
+    depth = 0
+    pending_blank = False
+    for idx, raw in enumerate(lines):
+        lineno = idx + 1
+        ops = by_line.get(lineno, [])
+
+        if lineno in verbatim:
+            if pending_blank:
+                out.append("")
+                pending_blank = False
+            out.append(raw)
+            depth = apply_depth(ops, depth)
+            continue
+
+        s = raw.strip()
+        if s == "":
+            pending_blank = bool(out)  # drop leading blanks
+            continue
+
+        # flush at most one blank — unless the next code line closes a
+        # brace (gofmt drops blanks before `}`)
+        if pending_blank and not s.startswith("}"):
+            out.append("")
+        pending_blank = False
+
+        leading_closers = 0
+        for op in ops:
+            if op is not None and op in ")]}":
+                leading_closers += 1
+            else:
+                break
+        indent = max(0, depth - leading_closers)
+        if s.startswith(("case ", "default:")) or s == "default:":
+            indent = max(0, indent - 1)
+        out.append("\t" * indent + s)
+        depth = apply_depth(ops, depth)
+
+    return "\n".join(out).rstrip("\n") + "\n"

@@ -3,14 +3,13 @@
 Covers the token classes that matter for import analysis and the static
 compile gate: comments (line + block), interpreted strings (with escape
 handling), raw strings, rune literals, identifiers/keywords, numbers,
-and operators/punctuation.  Semicolon insertion is NOT modeled — the
-consumers below work on token streams where newlines are preserved as
-NEWLINE tokens so they can reason about line structure when needed.
+and operators/punctuation.  Semicolon insertion is NOT modeled; line
+numbers are carried on each token for line-oriented consumers.
 """
 
 from __future__ import annotations
 
-from dataclasses import dataclass
+from typing import NamedTuple
 
 KEYWORDS = frozenset(
     """break case chan const continue default defer else fallthrough for
@@ -43,9 +42,8 @@ _TOKEN_RE = re.compile(
 _BAD_OPEN = re.compile(r'`[^`]*\Z|/\*(?:[^*]|\*(?!/))*\Z|"(?:\\.|[^"\\\n])*\Z')
 
 
-@dataclass(frozen=True)
-class Token:
-    kind: str  # IDENT KEYWORD STRING RAW_STRING RUNE NUMBER OP COMMENT NEWLINE
+class Token(NamedTuple):
+    kind: str  # IDENT KEYWORD STRING RAW_STRING RUNE NUMBER OP COMMENT
     text: str
     line: int  # 1-based
     col: int  # 0-based
@@ -80,7 +78,6 @@ def tokenize(src: str, keep_comments: bool = True) -> list[Token]:
         if kind == "WS":
             continue
         if kind == "NEWLINE":
-            append(Token("NEWLINE", "\n", line, m.start() - bol))
             line += 1
             bol = pos
             continue

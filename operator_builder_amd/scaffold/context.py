@@ -10,12 +10,15 @@ from ..workload.kinds import Resource, Workload
 
 def hash_fnv(s: str) -> str:
     """32-bit FNV-1a hash in hex — kubebuilder's ``hashFNV`` template
-    helper, used for the manager's LeaderElectionID."""
+    helper, used for the manager's LeaderElectionID.
+
+    Go renders ``fmt.Sprintf("%x", hasher.Sum(nil))`` over the 4-byte
+    sum, which is zero-padded to 8 hex chars — hence ``08x``."""
     h = 0x811C9DC5
     for byte in s.encode("utf-8"):
         h ^= byte
         h = (h * 0x01000193) & 0xFFFFFFFF
-    return format(h, "x")
+    return format(h, "08x")
 
 
 @dataclass

@@ -381,6 +381,7 @@ func Sample(requiredOnly bool) string {{
 
 \tfor _, f := range CreateFuncs {{
 \t\t{create_call}
+
 \t\tif err != nil {{
 \t\t\treturn nil, err
 \t\t}}
@@ -535,7 +536,10 @@ def _sample_const(ctx: Context, builder: Workload, required_only: bool) -> str:
     if not builder.is_cluster_scoped():
         lines.append("  namespace: default")
     spec = builder.get_api_spec_fields().generate_sample_spec(required_only)
-    return "\n".join(lines) + "\n" + spec.rstrip("\n")
+    # the reference embeds the sample via fmt.Sprintf of SampleTemplate
+    # whose GenerateSampleSpec output ends with a newline, so the raw
+    # string const closes on its own line (parity oracle verified)
+    return "\n".join(lines) + "\n" + spec.rstrip("\n") + "\n"
 
 
 def definition(ctx: Context, builder: Workload, manifest: Manifest) -> File:
@@ -584,7 +588,6 @@ func {child.create_func_name()}(
 \tparent *{res.import_alias}.{res.kind},
 {col_param}) ([]client.Object, error) {{
 {include_code}\tresourceObjs := []client.Object{{}}
-
 {source_code}
 
 {set_namespace}\tresourceObjs = append(resourceObjs, resourceObj)

@@ -198,19 +198,16 @@ def main_updater(
 
 
 def go_mod(ctx: Context) -> File:
+    """Byte-faithful rendering of the reference's goModTemplate
+    (templates/gomod.go:55-66): text/template's `{{ range }}` loop
+    leaves a tab-only separator line between entries and `{{ end -}}`
+    leaves the closing paren indented — go.mod is not gofmt'ed, so the
+    real output keeps that shape (verified by the parity oracle)."""
     deps = "".join(
-        f'\t"{name}" {version}\n\t'
+        f'\n\t"{name}" {version}\n\t'
         for name, version in sorted(GO_MOD_DEPENDENCIES.items())
     )
-    content = f"""
-module {ctx.repo}
-
-go 1.15
-
-require (
-\t{deps.strip()}
-)
-"""
+    content = f"\nmodule {ctx.repo}\n\ngo 1.15\n\nrequire (\n\t{deps})\n"
     return File("go.mod", content, IfExists.OVERWRITE)
 
 

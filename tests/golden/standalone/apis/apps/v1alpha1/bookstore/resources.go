@@ -46,7 +46,8 @@ spec:
     label: "bookstore"
   service:
     name: "bookstore"
-    targetPort: 8080`
+    targetPort: 8080
+`
 
 // sampleBookStoreRequired is a sample containing only required fields
 const sampleBookStoreRequired = `apiVersion: apps.example.com/v1alpha1
@@ -62,7 +63,8 @@ spec:
           required:
             port: 80
   service:
-    targetPort: 8080`
+    targetPort: 8080
+`
 
 // Sample returns the sample manifest for this custom resource.
 func Sample(requiredOnly bool) string {
@@ -80,6 +82,7 @@ func Generate(workloadObj appsv1alpha1.BookStore) ([]client.Object, error) {
 
 	for _, f := range CreateFuncs {
 		resources, err := f(&workloadObj)
+
 		if err != nil {
 			return nil, err
 		}

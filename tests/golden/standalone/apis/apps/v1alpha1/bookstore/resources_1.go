@@ -32,7 +32,6 @@ func CreateDeploymentBookstoreDeploy(
 	parent *appsv1alpha1.BookStore,
 ) ([]client.Object, error) {
 	resourceObjs := []client.Object{}
-
 	var resourceObj = &unstructured.Unstructured{
 		Object: map[string]interface{}{
 			"apiVersion": "apps/v1",
@@ -87,7 +86,6 @@ func CreateIngressBookstoreIng(
 	parent *appsv1alpha1.BookStore,
 ) ([]client.Object, error) {
 	resourceObjs := []client.Object{}
-
 	var resourceObj = &unstructured.Unstructured{
 		Object: map[string]interface{}{
 			"apiVersion": "networking.k8s.io/v1",
@@ -138,7 +136,6 @@ func CreateServiceServiceNameSvc(
 	parent *appsv1alpha1.BookStore,
 ) ([]client.Object, error) {
 	resourceObjs := []client.Object{}
-
 	var resourceObj = &unstructured.Unstructured{
 		Object: map[string]interface{}{
 			"kind": "Service",
@@ -181,7 +178,6 @@ func CreateRoleBookstoreRole(
 	parent *appsv1alpha1.BookStore,
 ) ([]client.Object, error) {
 	resourceObjs := []client.Object{}
-
 	var resourceObj = &unstructured.Unstructured{
 		Object: map[string]interface{}{
 			"apiVersion": "rbac.authorization.k8s.io/v1",

@@ -35,73 +35,65 @@ type BookStoreSpec struct {
 	// INSERT ADDITIONAL SPEC FIELDS - desired state of cluster
 	// Important: Run "make" to regenerate code after modifying this file
 
-// +kubebuilder:validation:Optional
-Bookstore BookStoreSpecBookstore `json:"bookstore,omitempty"`
+	// +kubebuilder:validation:Optional
+	Bookstore BookStoreSpecBookstore `json:"bookstore,omitempty"`
 
-// +kubebuilder:validation:Optional
-App BookStoreSpecApp `json:"app,omitempty"`
+	// +kubebuilder:validation:Optional
+	App BookStoreSpecApp `json:"app,omitempty"`
 
-// +kubebuilder:validation:Optional
-Service BookStoreSpecService `json:"service,omitempty"`
-
+	// +kubebuilder:validation:Optional
+	Service BookStoreSpecService `json:"service,omitempty"`
 }
 
 type BookStoreSpecBookstore struct{
-// +kubebuilder:validation:Optional
-Deeply BookStoreSpecBookstoreDeeply `json:"deeply,omitempty"`
-
+	// +kubebuilder:validation:Optional
+	Deeply BookStoreSpecBookstoreDeeply `json:"deeply,omitempty"`
 }
 
 type BookStoreSpecBookstoreDeeply struct{
-// +kubebuilder:validation:Optional
-Nested BookStoreSpecBookstoreDeeplyNested `json:"nested,omitempty"`
-
+	// +kubebuilder:validation:Optional
+	Nested BookStoreSpecBookstoreDeeplyNested `json:"nested,omitempty"`
 }
 
 type BookStoreSpecBookstoreDeeplyNested struct{
-// +kubebuilder:validation:Optional
-Path BookStoreSpecBookstoreDeeplyNestedPath `json:"path,omitempty"`
+	// +kubebuilder:validation:Optional
+	Path BookStoreSpecBookstoreDeeplyNestedPath `json:"path,omitempty"`
 
-// +kubebuilder:default="nginx:1.17"
-// +kubebuilder:validation:Optional
-// (Default: "nginx:1.17")
-// Defines the book store image
-Image string `json:"image,omitempty"`
-
+	// +kubebuilder:default="nginx:1.17"
+	// +kubebuilder:validation:Optional
+	// (Default: "nginx:1.17")
+	// Defines the book store image
+	Image string `json:"image,omitempty"`
 }
 
 type BookStoreSpecBookstoreDeeplyNestedPath struct{
-// +kubebuilder:default=2
-// +kubebuilder:validation:Optional
-// (Default: 2)
-Replicas int `json:"replicas,omitempty"`
+	// +kubebuilder:default=2
+	// +kubebuilder:validation:Optional
+	// (Default: 2)
+	Replicas int `json:"replicas,omitempty"`
 
-// +kubebuilder:validation:Optional
-Required BookStoreSpecBookstoreDeeplyNestedPathRequired `json:"required,omitempty"`
-
+	// +kubebuilder:validation:Optional
+	Required BookStoreSpecBookstoreDeeplyNestedPathRequired `json:"required,omitempty"`
 }
 
 type BookStoreSpecBookstoreDeeplyNestedPathRequired struct{
-Port int `json:"port,omitempty"`
-
+	Port int `json:"port,omitempty"`
 }
 
 type BookStoreSpecApp struct{
-// +kubebuilder:default="bookstore"
-// +kubebuilder:validation:Optional
-// (Default: "bookstore")
-Label string `json:"label,omitempty"`
-
+	// +kubebuilder:default="bookstore"
+	// +kubebuilder:validation:Optional
+	// (Default: "bookstore")
+	Label string `json:"label,omitempty"`
 }
 
 type BookStoreSpecService struct{
-// +kubebuilder:default="bookstore"
-// +kubebuilder:validation:Optional
-// (Default: "bookstore")
-Name string `json:"name,omitempty"`
+	// +kubebuilder:default="bookstore"
+	// +kubebuilder:validation:Optional
+	// (Default: "bookstore")
+	Name string `json:"name,omitempty"`
 
-TargetPort int `json:"targetPort,omitempty"`
-
+	TargetPort int `json:"targetPort,omitempty"`
 }
 
 // BookStoreStatus defines the observed state of BookStore.

@@ -208,9 +208,9 @@ func (r *BookStoreReconciler) SetupWithManager(mgr ctrl.Manager) error {
 	r.InitializePhases()
 
 	baseController, err := ctrl.NewControllerManagedBy(mgr).
-		WithEventFilter(predicates.WorkloadPredicates()).
-		For(&appsv1alpha1.BookStore{}).
-		Build(r)
+	WithEventFilter(predicates.WorkloadPredicates()).
+	For(&appsv1alpha1.BookStore{}).
+	Build(r)
 	if err != nil {
 		return fmt.Errorf("unable to setup controller, %w", err)
 	}

@@ -77,17 +77,17 @@ func main() {
 	// only print a given warning the first time we receive it
 	rest.SetDefaultWarningHandler(
 		rest.NewWarningWriter(os.Stderr, rest.WarningWriterOptions{
-			Deduplicate: true,
+				Deduplicate: true,
 		}),
 	)
 
 	mgr, err := ctrl.NewManager(ctrl.GetConfigOrDie(), ctrl.Options{
-		Scheme:                 scheme,
-		MetricsBindAddress:     metricsAddr,
-		Port:                   9443,
-		HealthProbeBindAddress: probeAddr,
-		LeaderElection:         enableLeaderElection,
-		LeaderElectionID:       "e3ca3705.example.com",
+			Scheme:                 scheme,
+			MetricsBindAddress:     metricsAddr,
+			Port:                   9443,
+			HealthProbeBindAddress: probeAddr,
+			LeaderElection:         enableLeaderElection,
+			LeaderElectionID:       "e3ca3705.example.com",
 	})
 	if err != nil {
 		setupLog.Error(err, "unable to start manager")

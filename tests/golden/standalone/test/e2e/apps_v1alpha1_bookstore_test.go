@@ -1,4 +1,3 @@
-//go:build e2e_test
 // +build e2e_test
 
 /*
@@ -35,6 +34,7 @@ import (
 // appsv1alpha1BookStore tests
 //
 func appsv1alpha1BookStoreChildrenFuncs(tester *E2ETest) error {
+	// TODO: need to run r.GetResources(request) on the reconciler to get the mutated resources
 	if len(bookstore.CreateFuncs) == 0 {
 		return nil
 	}
@@ -70,13 +70,21 @@ func (tester *E2ETest) appsv1alpha1BookStoreTest(testSuite *E2EComponentTestSuit
 	tester.suiteConfig = &testSuite.suiteConfig
 	require.NoErrorf(testSuite.T(), tester.setup(), "failed to setup test")
 
-	// create the custom resource and wait for its children to be ready
+	// create the custom resource
 	require.NoErrorf(testSuite.T(), testCreateCustomResource(tester), "failed to create custom resource")
 
-	// delete a whitelisted child and wait for the controller to restore it
+	// test the deletion of a child object
 	require.NoErrorf(testSuite.T(), testDeleteChildResource(tester), "failed to reconcile deletion of a child resource")
 
-	// verify the controller logged no errors for this workload
+	// test the update of a child object
+	// TODO: need immutable fields so that we can predict which managed fields we can modify to test reconciliation
+	// see https://github.com/vmware-tanzu-labs/operator-builder/issues/67
+
+	// test the update of a parent object
+	// TODO: need immutable fields so that we can predict which managed fields we can modify to test reconciliation
+	// see https://github.com/vmware-tanzu-labs/operator-builder/issues/67
+
+	// test that controller logs do not contain errors
 	if os.Getenv("DEPLOY_IN_CLUSTER") == "true" {
 		require.NoErrorf(testSuite.T(), testControllerLogsNoErrors(tester.suiteConfig, tester.logSyntax), "found errors in controller logs")
 	}

@@ -23,13 +23,11 @@ def test_tokenize_basic_kinds():
     assert kinds == [
         "KEYWORD",
         "IDENT",
-        "NEWLINE",
         "KEYWORD",
         "IDENT",
         "OP",
         "STRING",
         "COMMENT",
-        "NEWLINE",
     ]
 
 
