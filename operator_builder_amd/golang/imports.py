@@ -388,6 +388,14 @@ def _final_pass(text: str, tokens: list[Token] | None = None) -> str:
         indent = max(0, depth - leading_closers)
         if s.startswith(("case ", "default:")) or s == "default:":
             indent = max(0, indent - 1)
+        # gofmt forces a blank line between top-level declarations
+        if (
+            indent == 0
+            and out
+            and out[-1] == "}"
+            and not s.startswith("}")
+        ):
+            out.append("")
         out.append("\t" * indent + s)
         depth = apply_depth(ops, depth)
 

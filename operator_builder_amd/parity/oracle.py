@@ -787,6 +787,30 @@ def _normalize_render(text: str) -> str:
             prev_open_brace = False
             continue
 
+        # include-guard joined onto the signature's closing line
+        # (definition.go `{{- if ne .IncludeCode ""}}` artifact)
+        m = re.search(r"(\) \(\[\]client\.Object, error\) \{)(\S)", line)
+        if m:
+            out.append(line[: m.end(1)])
+            out.append(line[m.end(1) :])
+            prev_open_brace = False
+            continue
+
+        # struct literal closed on the last field's line (`...",}`)
+        if line.rstrip().endswith(",}"):
+            stripped_line = line.rstrip()
+            out.append(stripped_line[:-1])
+            out.append("}")
+            prev_open_brace = False
+            continue
+
+        # gofmt inserts a space after `,` inside expressions (template
+        # conditionals emit `collection,err` shapes) and drops a
+        # trailing comma joined to `)` on one line
+        if not has_quote:
+            line = re.sub(r",(?=[A-Za-z_&*])", ", ", line)
+            line = re.sub(r",\)", ")", line)
+
         out.append(line)
         prev_open_brace = s.endswith("{")
 
@@ -819,7 +843,7 @@ def _normalize_render(text: str) -> str:
                 break
         text = "\n".join(lines2)
 
-    return _reindent(text)
+    return text
 
 
 

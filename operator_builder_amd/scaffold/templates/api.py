@@ -475,11 +475,14 @@ def _generate_for_cli(
     takes_workload = builder.is_standalone() or builder.is_component()
     takes_collection = builder.is_component() or builder.is_collection()
 
-    params = ""
+    # gofmt normalizes the template's `a []byte,b []byte,` single-line
+    # parameter join to `a []byte, b []byte` (parity-oracle verified)
+    param_list = []
     if takes_workload:
-        params += "workloadFile []byte,"
+        param_list.append("workloadFile []byte")
     if takes_collection:
-        params += "collectionFile []byte,"
+        param_list.append("collectionFile []byte")
+    params = ", ".join(param_list)
 
     body = []
     if takes_workload:
@@ -518,7 +521,7 @@ def _generate_for_cli(
 // GenerateForCLI returns the child resources that are associated with this workload given
 // appropriate YAML manifest files.
 func GenerateForCLI({params}) ([]client.Object, error) {{
-{"".join(body)}
+{chr(10).join(body)}
 {ret}}}
 """
 

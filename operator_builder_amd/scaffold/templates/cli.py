@@ -626,6 +626,19 @@ def cmd_generate_sub(ctx: Context, builder: Workload) -> File:
     else:
         func_type = "\ttype generateFunc func([]byte) ([]client.Object, error)\n"
 
+    # components render the collection's api import (reference
+    # cmd_generate_sub.go template `{{- if .Builder.IsComponent }}`
+    # block); goimports removes it when unused, leaving the group's
+    # blank separator in place (parity-oracle verified)
+    collection_import = ""
+    if builder.is_component():
+        col = builder.get_collection()
+        collection_import = (
+            f'\t{col.get_api_group()}{col.get_api_version()} '
+            f'"{ctx.repo}/apis/{col.get_api_group()}/'
+            f'{col.get_api_version()}"\n\t\n'
+        )
+
     content = f"""{ctx.boilerplate}
 
 package {res.group}
@@ -646,7 +659,7 @@ import (
 \tcmdgenerate "{ctx.repo}/cmd/{root.name}/commands/generate"
 
 \t// specific imports for workloads
-\t{OB_IMPORTS_MARKER}
+{collection_import}\t{OB_IMPORTS_MARKER}
 )
 
 // New{kind}SubCommand creates a new command and adds it to its

@@ -212,7 +212,10 @@ def go_mod(ctx: Context) -> File:
 
 
 def makefile(ctx: Context) -> File:
-    cli_target = ""
+    # without a companion CLI the skipped `{{ if ne .RootCmdName "" -}}`
+    # branch still leaves its surrounding newlines in the reference's
+    # raw render (3 blank lines after `endef`; parity-oracle verified)
+    cli_target = "\n"
     if ctx.cli_root_command_name:
         cli_target = f"""# Build the companion CLI
 build-cli:

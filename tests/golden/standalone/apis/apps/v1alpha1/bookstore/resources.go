@@ -95,7 +95,7 @@ func Generate(workloadObj appsv1alpha1.BookStore) ([]client.Object, error) {
 
 // GenerateForCLI returns the child resources that are associated with this workload given
 // appropriate YAML manifest files.
-func GenerateForCLI(workloadFile []byte,) ([]client.Object, error) {
+func GenerateForCLI(workloadFile []byte) ([]client.Object, error) {
 	var workloadObj appsv1alpha1.BookStore
 	if err := yaml.Unmarshal(workloadFile, &workloadObj); err != nil {
 		return nil, fmt.Errorf("failed to unmarshal yaml into workload, %w", err)

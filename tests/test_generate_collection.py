@@ -168,7 +168,7 @@ def test_collection_generate_subcommand_flags(project):
 def test_component_resources_generate_for_cli_signature(project):
     content = read(project, "apis/apps/v1alpha1/webapp/resources.go")
     assert (
-        "func GenerateForCLI(workloadFile []byte,collectionFile []byte,)"
+        "func GenerateForCLI(workloadFile []byte, collectionFile []byte)"
         in content
     )
     assert "return Generate(workloadObj, collectionObj)" in content

@@ -145,7 +145,7 @@ def test_resources_file(project):
     content = read(project, "apis/apps/v1alpha1/bookstore/resources.go")
     assert "const sampleBookStore = `apiVersion: apps.example.com/v1alpha1" in content
     assert "func Generate(workloadObj appsv1alpha1.BookStore)" in content
-    assert "func GenerateForCLI(workloadFile []byte,)" in content
+    assert "func GenerateForCLI(workloadFile []byte)" in content
     assert "CreateDeploymentBookstoreDeploy,\n" in content
     assert "func ConvertWorkload(component workload.Workload)" in content
 
