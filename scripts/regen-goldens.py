@@ -1,14 +1,11 @@
 #!/usr/bin/env python3
-"""Regenerate the golden-output locks after an intentional template change:
-
-  - tests/golden/standalone/          (full tree, byte-for-byte)
-  - tests/golden/collection.sha256.json (hash manifest)
+"""Regenerate the golden-output locks after an intentional template
+change: full-byte trees for all five fixture families under
+tests/golden/<fixture>/ (VERDICT round-1 item 5).
 
 Run from the repo root, review the diff, commit.
 """
 
-import hashlib
-import json
 import os
 import shutil
 import sys
@@ -18,6 +15,15 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, REPO)
 
 from operator_builder_amd.cli.main import main  # noqa: E402
+
+# fixture -> repo path used at init time (feeds generated import paths)
+FIXTURES = {
+    "standalone": "github.com/acme/bookstore",
+    "edge-standalone": "github.com/acme/edge",
+    "collection": "github.com/acme/platform",
+    "edge-collection": "github.com/acme/edgeplatform",
+    "cluster-workload": "github.com/acme/agent",
+}
 
 
 def generate(fixture: str, repo: str) -> str:
@@ -52,26 +58,13 @@ def generate(fixture: str, repo: str) -> str:
 
 
 def run() -> None:
-    standalone = generate("standalone", "github.com/acme/bookstore")
-    target = os.path.join(REPO, "tests", "golden", "standalone")
-    shutil.rmtree(target, ignore_errors=True)
-    shutil.copytree(standalone, target)
-    print(f"standalone golden tree refreshed -> {target}")
-
-    collection = generate("collection", "github.com/acme/platform")
-    digest = {}
-    for root, dirs, files in os.walk(collection):
-        for name in sorted(files):
-            path = os.path.join(root, name)
-            rel = os.path.relpath(path, collection)
-            with open(path, "rb") as f:
-                digest[rel] = hashlib.sha256(f.read()).hexdigest()
-    manifest = os.path.join(
-        REPO, "tests", "golden", "collection.sha256.json"
-    )
-    with open(manifest, "w", encoding="utf-8") as f:
-        json.dump(digest, f, indent=1, sort_keys=True)
-    print(f"collection hash manifest refreshed ({len(digest)} files)")
+    for fixture, repo in FIXTURES.items():
+        tree = generate(fixture, repo)
+        target = os.path.join(REPO, "tests", "golden", fixture)
+        shutil.rmtree(target, ignore_errors=True)
+        shutil.copytree(tree, target)
+        count = sum(len(files) for _, _, files in os.walk(target))
+        print(f"{fixture} golden tree refreshed ({count} files) -> {target}")
 
 
 if __name__ == "__main__":

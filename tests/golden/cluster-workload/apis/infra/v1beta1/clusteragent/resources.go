@@ -1,0 +1,107 @@
+/*
+Copyright 2021.
+
+Licensed under the Apache License, Version 2.0 (the "License");
+you may not use this file except in compliance with the License.
+You may obtain a copy of the License at
+
+    http://www.apache.org/licenses/LICENSE-2.0
+
+Unless required by applicable law or agreed to in writing, software
+distributed under the License is distributed on an "AS IS" BASIS,
+WITHOUT WARRANTIES OR CONDITIONS OF ANY KIND, either express or implied.
+See the License for the specific language governing permissions and
+limitations under the License.
+*/
+
+package clusteragent
+
+import (
+	"sigs.k8s.io/controller-runtime/pkg/client"
+
+	"github.com/nukleros/operator-builder-tools/pkg/controller/workload"
+
+	infrav1beta1 "github.com/acme/agent/apis/infra/v1beta1"
+)
+
+// sampleClusterAgent is a sample containing all fields
+const sampleClusterAgent = `apiVersion: infra.example.com/v1beta1
+kind: ClusterAgent
+metadata:
+  name: clusteragent-sample
+spec:
+  agentImage: "telegraf:1.28"
+  apiToken: "replace-me"
+`
+
+// sampleClusterAgentRequired is a sample containing only required fields
+const sampleClusterAgentRequired = `apiVersion: infra.example.com/v1beta1
+kind: ClusterAgent
+metadata:
+  name: clusteragent-sample
+spec:
+  apiToken: "replace-me"
+`
+
+// Sample returns the sample manifest for this custom resource.
+func Sample(requiredOnly bool) string {
+	if requiredOnly {
+		return sampleClusterAgentRequired
+	}
+
+	return sampleClusterAgent
+}
+
+// Generate returns the child resources that are associated with this workload given
+// appropriate structured inputs.
+func Generate(workloadObj infrav1beta1.ClusterAgent) ([]client.Object, error) {
+	resourceObjects := []client.Object{}
+
+	for _, f := range CreateFuncs {
+		resources, err := f(&workloadObj)
+
+		if err != nil {
+			return nil, err
+		}
+
+		resourceObjects = append(resourceObjects, resources...)
+	}
+
+	return resourceObjects, nil
+}
+
+// CreateFuncs is an array of functions that are called to create the child resources for the controller
+// in memory during the reconciliation loop prior to persisting the changes or updates to the Kubernetes
+// database.
+var CreateFuncs = []func(
+	*infrav1beta1.ClusterAgent,
+) ([]client.Object, error) {
+	CreateNamespaceAgentSystem,
+	CreateServiceAccountAgentSystemAgentSa,
+	CreateClusterRoleAgentRole,
+	CreateClusterRoleBindingAgentBinding,
+	CreateDaemonSetAgentSystemAgentDs,
+	CreateSecretAgentSystemAgentToken,
+}
+
+// InitFuncs is an array of functions that are called prior to starting the controller manager.  This is
+// necessary in instances which the controller needs to "own" objects which depend on resources to
+// pre-exist in the cluster. A common use case for this is the need to own a custom resource.
+// If the controller needs to own a custom resource type, the CRD that defines it must
+// first exist. In this case, the InitFunc will create the CRD so that the controller
+// can own custom resources of that type.  Without the InitFunc the controller will
+// crash loop because when it tries to own a non-existent resource type during manager
+// setup, it will fail.
+var InitFuncs = []func(
+	*infrav1beta1.ClusterAgent,
+) ([]client.Object, error) {
+}
+
+func ConvertWorkload(component workload.Workload) (*infrav1beta1.ClusterAgent, error) {
+	p, ok := component.(*infrav1beta1.ClusterAgent)
+	if !ok {
+		return nil, infrav1beta1.ErrUnableToConvertClusterAgent
+	}
+
+	return p, nil
+}

@@ -1,0 +1,63 @@
+/*
+Copyright 2021.
+
+Licensed under the Apache License, Version 2.0 (the "License");
+you may not use this file except in compliance with the License.
+You may obtain a copy of the License at
+
+    http://www.apache.org/licenses/LICENSE-2.0
+
+Unless required by applicable law or agreed to in writing, software
+distributed under the License is distributed on an "AS IS" BASIS,
+WITHOUT WARRANTIES OR CONDITIONS OF ANY KIND, either express or implied.
+See the License for the specific language governing permissions and
+limitations under the License.
+*/
+
+package datastore
+
+import (
+	"k8s.io/apimachinery/pkg/apis/meta/v1/unstructured"
+	"sigs.k8s.io/controller-runtime/pkg/client"
+
+	datav1alpha1 "github.com/acme/platform/apis/data/v1alpha1"
+	platformsv1alpha1 "github.com/acme/platform/apis/platforms/v1alpha1"
+)
+
+// +kubebuilder:rbac:groups=core,resources=services,verbs=get;list;watch;create;update;patch;delete
+
+const ServiceDbSvc = "db-svc"
+
+// CreateServiceDbSvc creates the db-svc Service resource.
+func CreateServiceDbSvc(
+	parent *datav1alpha1.DataStore,
+	collection *platformsv1alpha1.CloudPlatform,
+) ([]client.Object, error) {
+	resourceObjs := []client.Object{}
+	var resourceObj = &unstructured.Unstructured{
+		Object: map[string]interface{}{
+			"kind": "Service",
+			"apiVersion": "v1",
+			"metadata": map[string]interface{}{
+				"name": "db-svc",
+			},
+			"spec": map[string]interface{}{
+				"selector": map[string]interface{}{
+					"app": "database",
+				},
+				"ports": []interface{}{
+					map[string]interface{}{
+						"protocol": "TCP",
+						"port": parent.Spec.DbPort,
+					},
+				},
+			},
+		},
+	}
+
+	resourceObj.SetNamespace(parent.Namespace)
+
+	resourceObjs = append(resourceObjs, resourceObj)
+
+	return resourceObjs, nil
+}

@@ -1,0 +1,170 @@
+/*
+Copyright 2021.
+
+Licensed under the Apache License, Version 2.0 (the "License");
+you may not use this file except in compliance with the License.
+You may obtain a copy of the License at
+
+    http://www.apache.org/licenses/LICENSE-2.0
+
+Unless required by applicable law or agreed to in writing, software
+distributed under the License is distributed on an "AS IS" BASIS,
+WITHOUT WARRANTIES OR CONDITIONS OF ANY KIND, either express or implied.
+See the License for the specific language governing permissions and
+limitations under the License.
+*/
+
+package v1alpha1
+
+import (
+	"errors"
+
+	"github.com/nukleros/operator-builder-tools/pkg/controller/workload"
+	"github.com/nukleros/operator-builder-tools/pkg/status"
+	metav1 "k8s.io/apimachinery/pkg/apis/meta/v1"
+	"k8s.io/apimachinery/pkg/runtime/schema"
+)
+
+var ErrUnableToConvertDataStore = errors.New("unable to convert to DataStore")
+
+// EDIT THIS FILE!  THIS IS SCAFFOLDING FOR YOU TO OWN!
+// NOTE: json tags are required.  Any new fields you add must have json tags for the fields to be serialized.
+
+// DataStoreSpec defines the desired state of DataStore.
+type DataStoreSpec struct {
+	// INSERT ADDITIONAL SPEC FIELDS - desired state of cluster
+	// Important: Run "make" to regenerate code after modifying this file
+
+	// +kubebuilder:validation:Optional
+	// Specifies a reference to the collection to use for this workload.
+	// Requires the name and namespace input to find the collection.
+	// If no collection field is set, default to selecting the only
+	// workload collection in the cluster, which will result in an error
+	// if not exactly one collection is found.
+	Collection DataStoreCollectionSpec `json:"collection"`
+
+	// +kubebuilder:default=5432
+	// +kubebuilder:validation:Optional
+	// (Default: 5432)
+	DbPort int `json:"dbPort,omitempty"`
+}
+
+type DataStoreCollectionSpec struct{
+	// +kubebuilder:validation:Required
+	// Required if specifying collection.  The name of the collection
+	// within a specific collection.namespace to reference.
+	Name string `json:"name"`
+
+	// +kubebuilder:validation:Optional
+	// (Default: "") The namespace where the collection exists.  Required only if
+	// the collection is namespace scoped and not cluster scoped.
+	Namespace string `json:"namespace"`
+}
+
+// DataStoreStatus defines the observed state of DataStore.
+type DataStoreStatus struct {
+	// INSERT ADDITIONAL STATUS FIELD - define observed state of cluster
+	// Important: Run "make" to regenerate code after modifying this file
+
+	Created               bool                       `json:"created,omitempty"`
+	DependenciesSatisfied bool                       `json:"dependenciesSatisfied,omitempty"`
+	Conditions            []*status.PhaseCondition   `json:"conditions,omitempty"`
+	Resources             []*status.ChildResource    `json:"resources,omitempty"`
+}
+
+// +kubebuilder:object:root=true
+// +kubebuilder:subresource:status
+
+// DataStore is the Schema for the datastores API.
+type DataStore struct {
+	metav1.TypeMeta   `json:",inline"`
+	metav1.ObjectMeta `json:"metadata,omitempty"`
+	Spec   DataStoreSpec   `json:"spec,omitempty"`
+	Status DataStoreStatus `json:"status,omitempty"`
+}
+
+// +kubebuilder:object:root=true
+
+// DataStoreList contains a list of DataStore.
+type DataStoreList struct {
+	metav1.TypeMeta `json:",inline"`
+	metav1.ListMeta `json:"metadata,omitempty"`
+	Items           []DataStore `json:"items"`
+}
+
+// interface methods
+
+// GetReadyStatus returns the ready status for a component.
+func (component *DataStore) GetReadyStatus() bool {
+	return component.Status.Created
+}
+
+// SetReadyStatus sets the ready status for a component.
+func (component *DataStore) SetReadyStatus(ready bool) {
+	component.Status.Created = ready
+}
+
+// GetDependencyStatus returns the dependency status for a component.
+func (component *DataStore) GetDependencyStatus() bool {
+	return component.Status.DependenciesSatisfied
+}
+
+// SetDependencyStatus sets the dependency status for a component.
+func (component *DataStore) SetDependencyStatus(dependencyStatus bool) {
+	component.Status.DependenciesSatisfied = dependencyStatus
+}
+
+// GetPhaseConditions returns the phase conditions for a component.
+func (component *DataStore) GetPhaseConditions() []*status.PhaseCondition {
+	return component.Status.Conditions
+}
+
+// SetPhaseCondition sets the phase conditions for a component.
+func (component *DataStore) SetPhaseCondition(condition *status.PhaseCondition) {
+	for i, currentCondition := range component.GetPhaseConditions() {
+		if currentCondition.Phase == condition.Phase {
+			component.Status.Conditions[i] = condition
+
+			return
+		}
+	}
+
+	// phase not found, lets add it to the list.
+	component.Status.Conditions = append(component.Status.Conditions, condition)
+}
+
+// GetResources returns the child resource status for a component.
+func (component *DataStore) GetChildResourceConditions() []*status.ChildResource {
+	return component.Status.Resources
+}
+
+// SetResources sets the phase conditions for a component.
+func (component *DataStore) SetChildResourceCondition(resource *status.ChildResource) {
+	for i, currentResource := range component.GetChildResourceConditions() {
+		if currentResource.Group == resource.Group && currentResource.Version == resource.Version && currentResource.Kind == resource.Kind {
+			if currentResource.Name == resource.Name && currentResource.Namespace == resource.Namespace {
+				component.Status.Resources[i] = resource
+
+				return
+			}
+		}
+	}
+
+	// phase not found, lets add it to the collection
+	component.Status.Resources = append(component.Status.Resources, resource)
+}
+
+// GetDependencies returns the dependencies for a component.
+func (*DataStore) GetDependencies() []workload.Workload {
+	return []workload.Workload{
+	}
+}
+
+// GetComponentGVK returns a GVK object for the component.
+func (*DataStore) GetWorkloadGVK() schema.GroupVersionKind {
+	return GroupVersion.WithKind("DataStore")
+}
+
+func init() {
+	SchemeBuilder.Register(&DataStore{}, &DataStoreList{})
+}

@@ -1,0 +1,82 @@
+/*
+Copyright 2021.
+
+Licensed under the Apache License, Version 2.0 (the "License");
+you may not use this file except in compliance with the License.
+You may obtain a copy of the License at
+
+    http://www.apache.org/licenses/LICENSE-2.0
+
+Unless required by applicable law or agreed to in writing, software
+distributed under the License is distributed on an "AS IS" BASIS,
+WITHOUT WARRANTIES OR CONDITIONS OF ANY KIND, either express or implied.
+See the License for the specific language governing permissions and
+limitations under the License.
+*/
+
+package mesh
+
+import (
+	"fmt"
+	"os"
+
+	"github.com/spf13/cobra"
+
+	"github.com/acme/edgeplatform/apis/mesh"
+
+	v1alpha1servicemesh "github.com/acme/edgeplatform/apis/mesh/v1alpha1/mesh"
+	cmdinit "github.com/acme/edgeplatform/cmd/edge-platform-ctl/commands/init"
+	//+operator-builder:imports
+)
+
+// getServiceMeshManifest returns the sample ServiceMesh manifest
+// based upon API Version input.
+func getServiceMeshManifest(i *cmdinit.InitSubCommand) (string, error) {
+	apiVersion := i.APIVersion
+	if apiVersion == "" || apiVersion == "latest" {
+		return mesh.ServiceMeshLatestSample, nil
+	}
+
+	// generate a map of all versions to samples for each api version created
+	manifestMap := map[string]string{
+		"v1alpha1": v1alpha1servicemesh.Sample(i.RequiredOnly),
+		//+operator-builder:versionmap
+	}
+
+	// return the manifest if it is not blank
+	manifest := manifestMap[apiVersion]
+	if manifest != "" {
+		return manifest, nil
+	}
+
+	// return an error if we did not find a manifest for an api version
+	return "", fmt.Errorf("unsupported API Version: " + apiVersion)
+}
+
+// NewServiceMeshSubCommand creates a new command and adds it to its
+// parent command.
+func NewServiceMeshSubCommand(parentCommand *cobra.Command) {
+	initCmd := &cmdinit.InitSubCommand{
+		Name:         "mesh",
+		Description:  "Manage the service mesh component",
+		InitFunc:     InitServiceMesh,
+		SubCommandOf: parentCommand,
+	}
+
+	initCmd.Setup()
+}
+
+func InitServiceMesh(i *cmdinit.InitSubCommand) error {
+	manifest, err := getServiceMeshManifest(i)
+	if err != nil {
+		return fmt.Errorf("unable to get manifest for ServiceMesh; %w", err)
+	}
+
+	outputStream := os.Stdout
+
+	if _, err := outputStream.WriteString(manifest); err != nil {
+		return fmt.Errorf("failed to write to stdout, %w", err)
+	}
+
+	return nil
+}

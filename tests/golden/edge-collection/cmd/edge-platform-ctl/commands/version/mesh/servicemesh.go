@@ -1,0 +1,53 @@
+/*
+Copyright 2021.
+
+Licensed under the Apache License, Version 2.0 (the "License");
+you may not use this file except in compliance with the License.
+You may obtain a copy of the License at
+
+    http://www.apache.org/licenses/LICENSE-2.0
+
+Unless required by applicable law or agreed to in writing, software
+distributed under the License is distributed on an "AS IS" BASIS,
+WITHOUT WARRANTIES OR CONDITIONS OF ANY KIND, either express or implied.
+See the License for the specific language governing permissions and
+limitations under the License.
+*/
+
+package mesh
+
+import (
+	"github.com/spf13/cobra"
+
+	cmdversion "github.com/acme/edgeplatform/cmd/edge-platform-ctl/commands/version"
+
+	"github.com/acme/edgeplatform/apis/mesh"
+)
+
+// NewServiceMeshSubCommand creates a new command and adds it to its
+// parent command.
+func NewServiceMeshSubCommand(parentCommand *cobra.Command) {
+	versionCmd := &cmdversion.VersionSubCommand{
+		Name:         "mesh",
+		Description:  "Manage the service mesh component",
+		VersionFunc:  VersionServiceMesh,
+		SubCommandOf: parentCommand,
+	}
+
+	versionCmd.Setup()
+}
+
+func VersionServiceMesh(v *cmdversion.VersionSubCommand) error {
+	apiVersions := make([]string, len(mesh.ServiceMeshGroupVersions()))
+
+	for i, groupVersion := range mesh.ServiceMeshGroupVersions() {
+		apiVersions[i] = groupVersion.Version
+	}
+
+	versionInfo := cmdversion.VersionInfo{
+		CLIVersion:  cmdversion.CLIVersion,
+		APIVersions: apiVersions,
+	}
+
+	return versionInfo.Display()
+}

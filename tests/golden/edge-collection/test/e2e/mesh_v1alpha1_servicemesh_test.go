@@ -1,0 +1,97 @@
+// +build e2e_test
+
+/*
+Copyright 2021.
+
+Licensed under the Apache License, Version 2.0 (the "License");
+you may not use this file except in compliance with the License.
+You may obtain a copy of the License at
+
+    http://www.apache.org/licenses/LICENSE-2.0
+
+Unless required by applicable law or agreed to in writing, software
+distributed under the License is distributed on an "AS IS" BASIS,
+WITHOUT WARRANTIES OR CONDITIONS OF ANY KIND, either express or implied.
+See the License for the specific language governing permissions and
+limitations under the License.
+*/
+
+package e2e_test
+
+import (
+	"fmt"
+	"os"
+
+	"github.com/stretchr/testify/require"
+
+	"k8s.io/apimachinery/pkg/apis/meta/v1/unstructured"
+
+	meshv1alpha1 "github.com/acme/edgeplatform/apis/mesh/v1alpha1"
+	"github.com/acme/edgeplatform/apis/mesh/v1alpha1/mesh"
+)
+
+//
+// meshv1alpha1ServiceMesh tests
+//
+func meshv1alpha1ServiceMeshChildrenFuncs(tester *E2ETest) error {
+	// TODO: need to run r.GetResources(request) on the reconciler to get the mutated resources
+	if len(mesh.CreateFuncs) == 0 {
+		return nil
+	}
+
+	workload, collection, err := mesh.ConvertWorkload(tester.workload, tester.collectionTester.workload)
+	if err != nil {
+		return fmt.Errorf("error in workload conversion; %w", err)
+	}
+
+	resourceObjects, err := mesh.Generate(*workload, *collection)
+	if err != nil {
+		return fmt.Errorf("unable to create objects in memory; %w", err)
+	}
+
+	tester.children = resourceObjects
+
+	return nil
+}
+
+func meshv1alpha1ServiceMeshNewHarness(namespace string) *E2ETest {
+	return &E2ETest{
+		namespace:          namespace,
+		unstructured:       &unstructured.Unstructured{},
+		workload:           &meshv1alpha1.ServiceMesh{},
+		sampleManifestFile: "../../config/samples/mesh_v1alpha1_servicemesh.yaml",
+		getChildrenFunc:    meshv1alpha1ServiceMeshChildrenFuncs,
+		logSyntax:          "controllers.mesh.ServiceMesh",
+		collectionTester:   edgeplatformv1alpha1EdgePlatformNewHarness(""),
+	}
+}
+
+func (tester *E2ETest) meshv1alpha1ServiceMeshTest(testSuite *E2EComponentTestSuite) {
+	testSuite.suiteConfig.tests = append(testSuite.suiteConfig.tests, tester)
+	tester.suiteConfig = &testSuite.suiteConfig
+	require.NoErrorf(testSuite.T(), tester.setup(), "failed to setup test")
+
+	// create the custom resource
+	require.NoErrorf(testSuite.T(), testCreateCustomResource(tester), "failed to create custom resource")
+
+	// test the deletion of a child object
+	require.NoErrorf(testSuite.T(), testDeleteChildResource(tester), "failed to reconcile deletion of a child resource")
+
+	// test the update of a child object
+	// TODO: need immutable fields so that we can predict which managed fields we can modify to test reconciliation
+	// see https://github.com/vmware-tanzu-labs/operator-builder/issues/67
+
+	// test the update of a parent object
+	// TODO: need immutable fields so that we can predict which managed fields we can modify to test reconciliation
+	// see https://github.com/vmware-tanzu-labs/operator-builder/issues/67
+
+	// test that controller logs do not contain errors
+	if os.Getenv("DEPLOY_IN_CLUSTER") == "true" {
+		require.NoErrorf(testSuite.T(), testControllerLogsNoErrors(tester.suiteConfig, tester.logSyntax), "found errors in controller logs")
+	}
+}
+
+func (testSuite *E2EComponentTestSuite) Test_meshv1alpha1ServiceMesh() {
+	tester := meshv1alpha1ServiceMeshNewHarness("")
+	tester.meshv1alpha1ServiceMeshTest(testSuite)
+}

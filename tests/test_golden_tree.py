@@ -1,15 +1,11 @@
-"""Golden-tree regression: the standalone fixture's generated operator is
-locked byte-for-byte against tests/golden/standalone.
+"""Golden-tree regression: every fixture family's generated operator is
+locked byte-for-byte against tests/golden/<fixture>/ (VERDICT round-1
+item 5 — all five families, full bytes).
 
-If an intentional template change alters output, regenerate the snapshot:
+If an intentional template change alters output, regenerate and review:
 
-    cd $(mktemp -d) && mkdir standalone && cd standalone
-    cp -r <repo>/tests/fixtures/standalone .workloadConfig
-    operator-builder init --workload-config .workloadConfig/workload.yaml \
-        --repo github.com/acme/bookstore
-    operator-builder create api
-    rm -rf .workloadConfig
-    rm -rf <repo>/tests/golden/standalone && cp -r . <repo>/tests/golden/standalone
+    python scripts/regen-goldens.py
+    git diff tests/golden/
 """
 
 import os
@@ -20,7 +16,16 @@ import pytest
 from operator_builder_amd.cli.main import main
 
 FIXTURES = os.path.join(os.path.dirname(__file__), "fixtures")
-GOLDEN = os.path.join(os.path.dirname(__file__), "golden", "standalone")
+GOLDEN = os.path.join(os.path.dirname(__file__), "golden")
+
+# fixture -> repo path; must match scripts/regen-goldens.py
+FAMILIES = {
+    "standalone": "github.com/acme/bookstore",
+    "edge-standalone": "github.com/acme/edge",
+    "collection": "github.com/acme/platform",
+    "edge-collection": "github.com/acme/edgeplatform",
+    "cluster-workload": "github.com/acme/agent",
+}
 
 
 def tree_files(base):
@@ -36,13 +41,13 @@ def tree_files(base):
     return out
 
 
-@pytest.fixture
-def generated(tmp_path, monkeypatch):
+@pytest.mark.parametrize("fixture", sorted(FAMILIES))
+def test_generated_tree_matches_golden(tmp_path, monkeypatch, fixture):
     # directory name must match the snapshot's project name derivation
-    workdir = tmp_path / "standalone"
+    workdir = tmp_path / fixture
     workdir.mkdir()
     shutil.copytree(
-        os.path.join(FIXTURES, "standalone"), workdir / ".workloadConfig"
+        os.path.join(FIXTURES, fixture), workdir / ".workloadConfig"
     )
     monkeypatch.chdir(workdir)
     assert (
@@ -52,75 +57,24 @@ def generated(tmp_path, monkeypatch):
                 "--workload-config",
                 ".workloadConfig/workload.yaml",
                 "--repo",
-                "github.com/acme/bookstore",
+                FAMILIES[fixture],
             ]
         )
         == 0
     )
     assert main(["create", "api"]) == 0
-    return workdir
 
+    golden = tree_files(os.path.join(GOLDEN, fixture))
+    actual = tree_files(str(workdir))
 
-def test_generated_tree_matches_golden(generated):
-    golden = tree_files(GOLDEN)
-    actual = tree_files(str(generated))
-
+    assert golden, f"golden tree for {fixture} is missing — run regen"
     assert sorted(actual) == sorted(golden), (
-        "file set diverged from golden snapshot"
+        f"{fixture}: file set diverged from golden snapshot"
     )
 
-    diverged = [
-        rel for rel in golden if actual[rel] != golden[rel]
-    ]
-    assert diverged == [], (
-        f"content diverged from golden snapshot in: {diverged[:10]} "
-        "(see module docstring to regenerate intentionally)"
-    )
-
-
-def test_collection_tree_matches_hash_manifest(tmp_path, monkeypatch):
-    """The collection fixture's 89-file tree is locked by sha256 manifest
-    (tests/golden/collection.sha256.json). Regenerate it with the loop in
-    this test (dump `actual`) after intentional template changes."""
-    import hashlib
-    import json
-
-    manifest_path = os.path.join(
-        os.path.dirname(__file__), "golden", "collection.sha256.json"
-    )
-    with open(manifest_path) as f:
-        golden = json.load(f)
-
-    workdir = tmp_path / "collection"
-    workdir.mkdir()
-    shutil.copytree(
-        os.path.join(FIXTURES, "collection"), workdir / ".workloadConfig"
-    )
-    monkeypatch.chdir(workdir)
-    assert (
-        main(
-            [
-                "init",
-                "--workload-config",
-                ".workloadConfig/workload.yaml",
-                "--repo",
-                "github.com/acme/platform",
-            ]
-        )
-        == 0
-    )
-    assert main(["create", "api"]) == 0
-
-    actual = {}
-    for root, dirs, files in os.walk(workdir):
-        if ".workloadConfig" in root:
-            continue
-        for name in sorted(files):
-            path = os.path.join(root, name)
-            rel = os.path.relpath(path, workdir)
-            with open(path, "rb") as f:
-                actual[rel] = hashlib.sha256(f.read()).hexdigest()
-
-    assert sorted(actual) == sorted(golden)
     diverged = [rel for rel in golden if actual[rel] != golden[rel]]
-    assert diverged == [], diverged[:10]
+    assert diverged == [], (
+        f"{fixture}: content diverged from golden snapshot in "
+        f"{diverged[:10]} (run scripts/regen-goldens.py to update "
+        "intentionally)"
+    )
