@@ -56,12 +56,39 @@ def _comment_lines(comment: str, prefix: str) -> list[str]:
 
 
 def _with_line_comment(line: str, node: Node) -> str:
-    if node.line_comment:
-        comment = node.line_comment.split("\n")[0].strip()
-        if not comment.startswith("#"):
-            comment = "# " + comment
-        return f"{line} {comment}"
+    trailing, _extra = _line_comment_parts(node)
+    if trailing:
+        return f"{line} {trailing}"
     return line
+
+
+def _line_comment_parts(node: Node) -> tuple[str, list[str]]:
+    """yaml.v3 renders a multi-line LineComment as: first line trailing
+    the value, remaining lines as full-width comment lines directly
+    below (at the value's indentation)."""
+    if not node.line_comment:
+        return "", []
+    raw_lines = node.line_comment.split("\n")
+    first = raw_lines[0].strip()
+    if first and not first.startswith("#"):
+        first = "# " + first
+    extra = []
+    for raw in raw_lines[1:]:
+        line = raw.strip()
+        if not line:
+            extra.append("#")
+            continue
+        if not line.startswith("#"):
+            line = "# " + line
+        extra.append(line)
+    return first, extra
+
+
+def _append_extra_line_comments(
+    lines: list[str], node: Node, prefix: str
+) -> None:
+    _first, extra = _line_comment_parts(node)
+    lines.extend(prefix + c for c in extra)
 
 
 # ---- scalars -----------------------------------------------------------
@@ -221,6 +248,55 @@ def _render_inline_value(node: Node) -> str:
     return _render_flow(node)
 
 
+# ---- plain-scalar line wrapping ------------------------------------------
+
+BEST_WIDTH = 80  # libyaml / yaml.v3 emitter best_width
+
+
+def _wrap_plain(line: str, prefix: str) -> list[str]:
+    """Wrap a line holding a plain scalar the way libyaml's
+    yaml_emitter_write_plain_scalar does: while writing, at each single
+    space (not adjacent to another space) reached when the column is
+    already past best_width, a line break replaces the space and the
+    scalar continues indented one step deeper.  Re-parsing folds the
+    break back to a single space, so the value is unchanged."""
+    if len(line) <= BEST_WIDTH or " " not in line.strip():
+        return [line]
+    cont_prefix = prefix + INDENT
+    out: list[str] = []
+    current = line
+    while len(current) > BEST_WIDTH:
+        # find the first breakable space past the width
+        pos = None
+        i = BEST_WIDTH
+        while i < len(current):
+            if (
+                current[i] == " "
+                and current[i - 1] != " "
+                and i + 1 < len(current)
+                and current[i + 1] != " "
+            ):
+                pos = i
+                break
+            i += 1
+        if pos is None:
+            break
+        out.append(current[:pos])
+        current = cont_prefix + current[pos + 1 :]
+    out.append(current)
+    return out
+
+
+def _wrappable(node: Node) -> bool:
+    return (
+        node.kind == SCALAR
+        and node.tag == TAG_STR
+        and node.style not in ("'", '"', "|", ">")
+        and _plain_safe(node.value)
+        and not node.line_comment
+    )
+
+
 # ---- block emission ----------------------------------------------------
 
 
@@ -259,10 +335,13 @@ def _emit_block_mapping(node: Node, prefix: str) -> list[str]:
             lines.extend(block[1:])
         elif _is_inline_value(value):
             line = f"{prefix}{key_text}: {_render_inline_value(value)}"
-            line = _with_line_comment(
-                line, value if value.line_comment else key
-            )
-            lines.append(line)
+            comment_node = value if value.line_comment else key
+            line = _with_line_comment(line, comment_node)
+            if _wrappable(value) and not comment_node.line_comment:
+                lines.extend(_wrap_plain(line, prefix))
+            else:
+                lines.append(line)
+                _append_extra_line_comments(lines, comment_node, prefix)
         else:
             line = _with_line_comment(f"{prefix}{key_text}:", key)
             lines.append(line)
@@ -295,7 +374,12 @@ def _emit_block_sequence(node: Node, prefix: str) -> list[str]:
             first = prefix + "-"
         if item.kind == SCALAR:
             first = _with_line_comment(first, item)
-        lines.append(first)
+        if item.kind == SCALAR and _wrappable(item):
+            lines.extend(_wrap_plain(first, prefix + "  "))
+        else:
+            lines.append(first)
+            if item.kind == SCALAR:
+                _append_extra_line_comments(lines, item, prefix)
         lines.extend(item_lines[1:])
 
         if item.foot_comment:
