@@ -31,6 +31,7 @@ from operator_builder_amd.workload import config as workload_config
 from operator_builder_amd.workload import subcommand
 
 REFERENCE_CASES = "/root/reference/test/cases"
+FIXTURES = os.path.join(os.path.dirname(__file__), "fixtures")
 
 pytestmark = pytest.mark.skipif(
     not reference_available(),
@@ -38,17 +39,7 @@ pytestmark = pytest.mark.skipif(
 )
 
 
-@pytest.mark.parametrize(
-    "case",
-    ["standalone", "edge-standalone", "collection", "edge-collection"],
-)
-def test_generated_tree_matches_reference_templates(tmp_path, case):
-    workdir = tmp_path / case
-    workdir.mkdir()
-    shutil.copytree(
-        os.path.join(REFERENCE_CASES, case, ".workloadConfig"),
-        workdir / ".workloadConfig",
-    )
+def _assert_parity(workdir, repo):
     cwd = os.getcwd()
     os.chdir(workdir)
     try:
@@ -59,7 +50,7 @@ def test_generated_tree_matches_reference_templates(tmp_path, case):
                     "--workload-config",
                     ".workloadConfig/workload.yaml",
                     "--repo",
-                    f"github.com/acme/{case.replace('-', '')}",
+                    repo,
                 ]
             )
             == 0
@@ -83,6 +74,42 @@ def test_generated_tree_matches_reference_templates(tmp_path, case):
         r.path: r.diff_lines for r in report if r.diff_lines != 0
     }
     assert not mismatched, (
-        f"{case}: {len(mismatched)}/{len(report)} files diverge from the "
+        f"{len(mismatched)}/{len(report)} files diverge from the "
         f"reference templates: {mismatched}"
     )
+
+
+@pytest.mark.parametrize(
+    "case",
+    ["standalone", "edge-standalone", "collection", "edge-collection"],
+)
+def test_reference_fixture_parity(tmp_path, case):
+    workdir = tmp_path / case
+    workdir.mkdir()
+    shutil.copytree(
+        os.path.join(REFERENCE_CASES, case, ".workloadConfig"),
+        workdir / ".workloadConfig",
+    )
+    _assert_parity(workdir, f"github.com/acme/{case.replace('-', '')}")
+
+
+@pytest.mark.parametrize(
+    "fixture",
+    [
+        "standalone",
+        "edge-standalone",
+        "collection",
+        "edge-collection",
+        "cluster-workload",
+    ],
+)
+def test_bundled_fixture_parity(tmp_path, fixture):
+    """The oracle also holds on the in-repo fixture families — notably
+    cluster-workload, which exercises the cluster-scoped template
+    branches the reference fixtures leave partially covered."""
+    workdir = tmp_path / fixture
+    workdir.mkdir()
+    shutil.copytree(
+        os.path.join(FIXTURES, fixture), workdir / ".workloadConfig"
+    )
+    _assert_parity(workdir, f"github.com/acme/{fixture.replace('-', '')}")
