@@ -80,7 +80,7 @@ class FieldMarker:
     def __str__(self) -> str:
         return (
             f"FieldMarker{{Name: {self.name} Type: {self.type} "
-            f"Description: {self.get_description()!r} "
+            f'Description: "{self.get_description()}" '
             f"Default: {self.default}}}"
         )
 
@@ -143,7 +143,7 @@ class CollectionFieldMarker(FieldMarker):
     def __str__(self) -> str:
         return (
             f"CollectionFieldMarker{{Name: {self.name} Type: {self.type} "
-            f"Description: {self.get_description()!r} "
+            f'Description: "{self.get_description()}" '
             f"Default: {self.default}}}"
         )
 
