@@ -25,6 +25,22 @@ class APIFieldError(OperatorBuilderError):
 ERR_OVERWRITE = "an attempt to overwrite existing value was made"
 
 
+def _go_fmt_value(value: Any) -> str:
+    """Go fmt %v rendering for the non-scalar values a manifest can
+    carry (reference api.go getSampleValue uses fmt.Sprintf("%v", ...)):
+    slices render as [a b c], maps as map[k:v]."""
+    if isinstance(value, bool):
+        return "true" if value else "false"
+    if isinstance(value, (list, tuple)):
+        return "[" + " ".join(_go_fmt_value(v) for v in value) + "]"
+    if isinstance(value, dict):
+        inner = " ".join(
+            f"{k}:{_go_fmt_value(v)}" for k, v in sorted(value.items())
+        )
+        return f"map[{inner}]"
+    return str(value)
+
+
 @dataclass
 class APIFields:
     name: str
@@ -133,7 +149,7 @@ class APIFields:
             return sample_val
         if isinstance(sample_val, bool):
             return "true" if sample_val else "false"
-        return f"{sample_val}"
+        return _go_fmt_value(sample_val)
 
     def set_sample(self, sample_val: Any) -> None:
         if self.type == FieldType.STRUCT:
