@@ -23,7 +23,7 @@ from ..license import (
     update_source_header,
 )
 from ..scaffold.context import Context
-from ..scaffold.project import Project, ProjectError
+from ..scaffold.project import Project
 from ..scaffold.scaffolder import (
     resource_for_workload,
     scaffold_api,

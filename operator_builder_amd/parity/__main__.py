@@ -7,7 +7,6 @@ summary.  With -v, prints unified diffs for mismatching files.
 """
 
 import difflib
-import os
 import sys
 
 from ..cli.main import _build_context

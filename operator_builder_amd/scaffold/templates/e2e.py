@@ -1,3 +1,4 @@
+# lint: allow-tabs (embedded Go template text)
 """E2E test templates for the generated operator.
 
 Parity targets: reference templates/test/e2e/e2e.go:22-875 (the common

@@ -4,7 +4,7 @@ example budget (CI runs are derandomized for stability; run this after
 substantive yamlast/marker changes)."""
 import sys, os
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
-from hypothesis import given, settings, strategies as st
+from hypothesis import given, settings
 import tests.test_yamlast_property as t1
 import tests.test_marker_fuzz as t2
 
