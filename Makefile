@@ -3,9 +3,9 @@
 
 PYTHON ?= python3
 
-.PHONY: all test test-unit test-func test-gpu bench smoke lint clean
+.PHONY: all test test-unit test-func test-gpu bench smoke lint parity release clean
 
-all: test
+all: lint test
 
 test:
 	$(PYTHON) -m pytest tests/ -q -m "not gpu"
@@ -30,6 +30,18 @@ bench:
 
 smoke:
 	$(PYTHON) __graft_entry__.py
+
+# AST-based lint gate (analog of the reference's golangci-lint step)
+lint:
+	$(PYTHON) scripts/lint.py
+
+# render the reference templates against a generated tree (diff==0 check)
+parity:
+	$(PYTHON) -m pytest tests/test_reference_parity.py -q
+
+# build distributables + shell completions (analog of .goreleaser.yml)
+release:
+	bash scripts/release.sh
 
 clean:
 	find . -name __pycache__ -type d -prune -exec rm -rf {} +

@@ -267,3 +267,19 @@ class TestInitConfigRoundTrip:
         )
         assert os.path.exists("PROJECT")
         assert os.path.exists("main.go")
+
+
+def test_lint_gate_clean():
+    """The AST lint gate (scripts/lint.py, golangci-lint analog) stays
+    clean over the package and scripts."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    result = subprocess.run(
+        [sys.executable, os.path.join(repo, "scripts", "lint.py")],
+        cwd=repo,
+        capture_output=True,
+        text=True,
+    )
+    assert result.returncode == 0, result.stdout + result.stderr
