@@ -189,3 +189,24 @@ def test_check_tree_duplicate_funcs_and_mixed_packages(tmp_path):
     (d / "b.go").write_text("package b\n\nfunc G() {}\n")
     issues = check_tree(str(tmp_path))
     assert any("mixed package names" in str(i) for i in issues)
+
+
+def test_format_go_idempotent_over_golden_trees():
+    """format_go must be a fixed point on its own output for every
+    generated .go file (otherwise repeated create api runs would churn
+    bytes)."""
+    import os
+
+    golden = os.path.join(os.path.dirname(__file__), "golden")
+    checked = 0
+    for root, _dirs, files in os.walk(golden):
+        for name in files:
+            if not name.endswith(".go"):
+                continue
+            path = os.path.join(root, name)
+            with open(path, encoding="utf-8") as f:
+                content = f.read()
+            once = format_go(content)
+            assert once == content, f"{path}: golden not a fixed point"
+            checked += 1
+    assert checked > 100
