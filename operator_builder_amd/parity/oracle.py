@@ -23,7 +23,6 @@ import re
 from dataclasses import dataclass
 
 from ..golang import format_go
-from ..golang.imports import _reindent
 from ..scaffold.context import Context
 from ..scaffold.machinery import Marker, insert_code_fragments
 from ..scaffold.scaffolder import resource_for_workload

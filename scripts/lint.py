@@ -109,7 +109,7 @@ class _Lint(ast.NodeVisitor):
 
         is_package_init = os.path.basename(self.path) == "__init__.py"
         for name, lineno in sorted(self.imported.items()):
-            if name.startswith("_"):
+            if name == "_":
                 continue
             if name in self.used or name in self.export_all:
                 continue
