@@ -21,7 +21,7 @@ import re
 from dataclasses import dataclass
 
 from .lexer import Token, tokenize, GoLexError
-from .imports import _IMPORT_LINE, _default_name
+from .imports import _IMPORT_LINE, _default_name, _used_identifiers
 
 # standard-library packages that generated code plausibly references;
 # used for "qualifier used but not imported" detection without type info
@@ -135,8 +135,6 @@ def check_file(
         seen_paths[ipath] = name
 
     # usage analysis
-    from .imports import _used_identifiers
-
     used = _used_identifiers(tokens)
 
     for name, (ipath, lineno) in imports.items():

@@ -18,6 +18,7 @@ documented in PARITY.md ("marker format").
 
 from __future__ import annotations
 
+import difflib
 import os
 import re
 from dataclasses import dataclass
@@ -875,8 +876,6 @@ def diff_report(
         if generated == oracle:
             report.append(FileDiff(path, 0))
         else:
-            import difflib
-
             diff = list(
                 difflib.unified_diff(
                     oracle.splitlines(),
