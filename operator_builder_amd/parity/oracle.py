@@ -136,10 +136,12 @@ def _render_cli_root(ctx: Context, workload: Workload) -> dict[str, str]:
     name = workload.get_root_command().name
     builder = builder_shape(workload)
 
-    # templates/cli/main.go:24-37
-    files[f"cmd/{name}/main.go"] = _tpl("cli/main.go", "cliMainTemplate").render(
-        {**base, "RootCmd": root_cmd}
-    )
+    # templates/cli/main.go:24-37 (SkipFile)
+    cli_main_path = f"cmd/{name}/main.go"
+    if cli_main_path not in files:
+        files[cli_main_path] = _tpl(
+            "cli/main.go", "cliMainTemplate"
+        ).render({**base, "RootCmd": root_cmd})
 
     # templates/cli/cmd_root.go:35-50 (markers cmd_root.go:78-81)
     files[f"cmd/{name}/commands/root.go"] = _tpl(
@@ -315,18 +317,25 @@ def oracle_render_workload(
         }
     )
 
-    # controller/phases.go:26-38
-    files[f"controllers/{group}/{to_file_name(kind)}_phases.go"] = _tpl(
-        "controller/phases.go", "phasesTemplate"
-    ).render({**base, "PackageName": pkg})
+    # controller/phases.go:26-38 (SkipFile)
+    phases_path = f"controllers/{group}/{to_file_name(kind)}_phases.go"
+    if phases_path not in files:
+        files[phases_path] = _tpl(
+            "controller/phases.go", "phasesTemplate"
+        ).render({**base, "PackageName": pkg})
 
     # int/dependencies/component.go:26-38, int/mutate/component.go
-    files[f"internal/dependencies/{to_file_name(kind)}.go"] = _tpl(
-        "int/dependencies/component.go", "componentTemplate"
-    ).render(base)
-    files[f"internal/mutate/{to_file_name(kind)}.go"] = _tpl(
-        "int/mutate/component.go", "componentTemplate"
-    ).render(base)
+    # (both SkipFile)
+    dep_path = f"internal/dependencies/{to_file_name(kind)}.go"
+    if dep_path not in files:
+        files[dep_path] = _tpl(
+            "int/dependencies/component.go", "componentTemplate"
+        ).render(base)
+    mut_path = f"internal/mutate/{to_file_name(kind)}.go"
+    if mut_path not in files:
+        files[mut_path] = _tpl(
+            "int/mutate/component.go", "componentTemplate"
+        ).render(base)
 
     # controller/controller_suitetest.go:30-55 + fragments :78-104
     suite_path = f"controllers/{group}/suite_test.go"
@@ -414,10 +423,12 @@ def oracle_render_workload(
         }
     )
 
-    # test/e2e/workloads.go:44-76 (+helpers :214-260)
-    files[
-        f"test/e2e/{group}_{version}_{kind_lower}_test.go"
-    ] = _tpl("test/e2e/workloads.go", "e2eWorkloadsTemplate").render(
+    # test/e2e/workloads.go:44-76 (+helpers :214-260; SkipFile)
+    e2e_path = f"test/e2e/{group}_{version}_{kind_lower}_test.go"
+    if e2e_path not in files:
+        files[e2e_path] = _tpl(
+            "test/e2e/workloads.go", "e2eWorkloadsTemplate"
+        ).render(
         {
             **base,
             "Builder": builder,
@@ -553,7 +564,8 @@ def _render_cli_sub(
     else:
         init_name, init_descr = sub_cmd_obj.name, sub_cmd_obj.description
     p = sub_path("init")
-    files[p] = _tpl(
+    if p not in files:  # SkipFile (matches this repo's cmd_init_sub)
+        files[p] = _tpl(
         "cli/cmd_init_sub.go",
         "cmdInitSub",
         (cli_tpl.OB_IMPORTS_MARKER, cli_tpl.OB_VERSIONMAP_MARKER),
@@ -601,7 +613,8 @@ def _render_cli_sub(
             gen_inputs = "workloadFile"
         col = workload.get_collection()
         p = sub_path("generate")
-        files[p] = _tpl(
+        if p not in files:  # SkipFile
+            files[p] = _tpl(
             "cli/cmd_generate_sub.go",
             "cmdGenerateSub",
             (
@@ -640,7 +653,8 @@ def _render_cli_sub(
     else:
         ver_name, ver_descr = sub_cmd_obj.name, sub_cmd_obj.description
     p = sub_path("version")
-    files[p] = _tpl("cli/cmd_version_sub.go", "cmdVersionSub").render(
+    if p not in files:  # SkipFile
+        files[p] = _tpl("cli/cmd_version_sub.go", "cmdVersionSub").render(
         {
             **base,
             "Builder": builder,
@@ -860,8 +874,22 @@ def diff_report(
     base_dir: str, ctx: Context, workload: Workload
 ) -> list[FileDiff]:
     """Render the oracle for a generated tree and diff every file."""
+    return diff_report_sequence(base_dir, [(ctx, workload)])
+
+
+def diff_report_sequence(
+    base_dir: str, runs: list[tuple[Context, Workload]]
+) -> list[FileDiff]:
+    """Replay a sequence of create-api runs (e.g. the documented API
+    version-upgrade workflow, docs/api-updates-upgrades.md) through the
+    oracle: the first run renders init + api files, later runs overwrite
+    OverwriteFile templates, skip SkipFile ones, and accumulate inserter
+    fragments — exactly like the real pipeline — then diff the result
+    against the on-disk tree."""
+    ctx, workload = runs[0]
     files = render_init_files(ctx, workload)
-    oracle_render_workload(files, ctx, workload)
+    for run_ctx, run_workload in runs:
+        oracle_render_workload(files, run_ctx, run_workload)
 
     report: list[FileDiff] = []
     for path, oracle in sorted(files.items()):

@@ -113,3 +113,73 @@ def test_bundled_fixture_parity(tmp_path, fixture):
         os.path.join(FIXTURES, fixture), workdir / ".workloadConfig"
     )
     _assert_parity(workdir, f"github.com/acme/{fixture.replace('-', '')}")
+
+
+def test_version_upgrade_sequence_parity(tmp_path):
+    """The documented version-upgrade workflow
+    (docs/api-updates-upgrades.md: bump spec.api.version, re-run
+    `create api --force`) replayed through the oracle: the accumulated
+    tree (both API versions, extended kind registry, CLI version maps,
+    main.go fragments) stays byte-identical to the reference templates."""
+    from operator_builder_amd.parity.oracle import diff_report_sequence
+
+    workdir = tmp_path / "upgrade"
+    workdir.mkdir()
+    shutil.copytree(
+        os.path.join(FIXTURES, "standalone"), workdir / ".workloadConfig"
+    )
+    cfg = workdir / ".workloadConfig" / "workload.yaml"
+
+    cwd = os.getcwd()
+    os.chdir(workdir)
+    try:
+        assert (
+            main(
+                [
+                    "init",
+                    "--workload-config",
+                    ".workloadConfig/workload.yaml",
+                    "--repo",
+                    "github.com/acme/upgrade",
+                ]
+            )
+            == 0
+        )
+        assert main(["create", "api"]) == 0
+
+        project = Project.load(".")
+        processor_v1 = workload_config.parse(str(cfg))
+        subcommand.create_api(processor_v1)
+        ctx_v1 = _build_context(".", project, processor_v1.workload)
+
+        # bump the API version and regenerate (documented workflow)
+        cfg.write_text(
+            cfg.read_text().replace("version: v1alpha1", "version: v1alpha2")
+        )
+        assert main(["create", "api", "--force"]) == 0
+
+        project = Project.load(".")
+        processor_v2 = workload_config.parse(str(cfg))
+        subcommand.create_api(processor_v2)
+        ctx_v2 = _build_context(".", project, processor_v2.workload)
+
+        report = diff_report_sequence(
+            ".",
+            [
+                (ctx_v1, processor_v1.workload),
+                (ctx_v2, processor_v2.workload),
+            ],
+        )
+    finally:
+        os.chdir(cwd)
+
+    bad = {
+        r.path: ("MISSING" if r.missing else r.diff_lines)
+        for r in report
+        if r.missing or r.diff_lines != 0
+    }
+    assert not bad, bad
+    # both versions' files must be part of the oracle-covered set
+    paths = {r.path for r in report}
+    assert any("v1alpha1" in p for p in paths)
+    assert any("v1alpha2" in p for p in paths)
