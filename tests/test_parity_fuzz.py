@@ -181,3 +181,145 @@ def test_random_workloads_stay_byte_identical(tmp_path_factory, setup):
         if r.missing or r.diff_lines != 0
     }
     assert not bad, f"diverged for config:\n{config}\n{manifest}\n{bad}"
+
+
+@st.composite
+def collection_setups(draw):
+    """A collection + 1-2 components with dependencies, collection
+    markers, and optionally a resource marker on a collectionField."""
+    col_group = draw(kinds_names)
+    col_kind = draw(kinds_names).capitalize() + "Platform"
+    n_components = draw(st.integers(min_value=1, max_value=2))
+    col_cluster = draw(st.booleans())
+
+    col_field = draw(names)
+    files = {}
+
+    comp_blocks = []
+    comp_names = []
+    for i in range(n_components):
+        comp_name = f"comp-{draw(names)}-{i}"
+        comp_names.append(comp_name)
+    for i, comp_name in enumerate(comp_names):
+        group = draw(kinds_names)
+        kind = draw(kinds_names).capitalize() + f"Part{i}"
+        dep_block = ""
+        if i > 0 and draw(st.booleans()):
+            dep_block = f"  dependencies:\n  - {comp_names[0]}\n"
+        use_resource_marker = draw(st.booleans())
+        rm_line = ""
+        if use_resource_marker:
+            rm_line = (
+                f"# +operator-builder:resource:collectionField="
+                f"{col_field},value=\"prod\",include\n"
+            )
+        field = draw(names)
+        files[f"{comp_name}/component.yaml"] = (
+            f"name: {comp_name}\n"
+            "kind: ComponentWorkload\n"
+            "spec:\n"
+            "  api:\n"
+            f"    group: {group}\n"
+            "    version: v1alpha1\n"
+            f"    kind: {kind}\n"
+            f"    clusterScoped: {'true' if draw(st.booleans()) else 'false'}\n"
+            "  companionCliSubcmd:\n"
+            f"    name: {draw(names)}\n"
+            "    description: Manage the component\n"
+            f"{dep_block}"
+            "  resources:\n"
+            "  - resources.yaml\n"
+        )
+        files[f"{comp_name}/resources.yaml"] = (
+            f"{rm_line}"
+            "apiVersion: v1\n"
+            "kind: ConfigMap\n"
+            "metadata:\n"
+            f"  name: {comp_name}-config\n"
+            "  namespace: default\n"
+            "data:\n"
+            f"  # +operator-builder:field:name={field},type=string,"
+            'default="x"\n'
+            '  key: "x"\n'
+        )
+        comp_blocks.append(f"  - {comp_name}/component.yaml")
+
+    files["workload.yaml"] = (
+        f"name: {draw(names)}-collection\n"
+        "kind: WorkloadCollection\n"
+        "spec:\n"
+        "  api:\n"
+        "    domain: example.com\n"
+        f"    group: {col_group}\n"
+        "    version: v1alpha1\n"
+        f"    kind: {col_kind}\n"
+        f"    clusterScoped: {'true' if col_cluster else 'false'}\n"
+        "  companionCliRootcmd:\n"
+        f"    name: {draw(names)}ctl\n"
+        "    description: Manage the platform\n"
+        "  resources:\n"
+        "  - settings.yaml\n"
+        "  componentFiles:\n" + "\n".join(comp_blocks) + "\n"
+    )
+    files["settings.yaml"] = (
+        "apiVersion: v1\n"
+        "kind: ConfigMap\n"
+        "metadata:\n"
+        "  name: platform-settings\n"
+        "  namespace: default\n"
+        "data:\n"
+        f"  # +operator-builder:collection:field:name={col_field},"
+        'type=string,default="prod"\n'
+        '  env: "prod"\n'
+    )
+    return files
+
+
+@settings(
+    max_examples=8,
+    deadline=None,
+    derandomize=True,
+    suppress_health_check=[HealthCheck.too_slow],
+)
+@given(collection_setups())
+def test_random_collections_stay_byte_identical(tmp_path_factory, files):
+    workdir = tmp_path_factory.mktemp("parityfuzzcol")
+    cfg_dir = workdir / ".workloadConfig"
+    cfg_dir.mkdir()
+    for rel, content in files.items():
+        dest = cfg_dir / rel
+        dest.parent.mkdir(parents=True, exist_ok=True)
+        dest.write_text(content)
+
+    cwd = os.getcwd()
+    os.chdir(workdir)
+    try:
+        assert (
+            main(
+                [
+                    "init",
+                    "--workload-config",
+                    ".workloadConfig/workload.yaml",
+                    "--repo",
+                    "github.com/fuzz/platform",
+                ]
+            )
+            == 0
+        ), files["workload.yaml"]
+        assert main(["create", "api"]) == 0, files["workload.yaml"]
+
+        project = Project.load(".")
+        processor = workload_config.parse(".workloadConfig/workload.yaml")
+        subcommand.create_api(processor)
+        ctx = _build_context(".", project, processor.workload)
+        report = diff_report(".", ctx, processor.workload)
+    finally:
+        os.chdir(cwd)
+        shutil.rmtree(workdir, ignore_errors=True)
+
+    bad = {
+        r.path: ("MISSING" if r.missing else r.diff_lines)
+        for r in report
+        if r.missing or r.diff_lines != 0
+    }
+    assert not bad, f"diverged for:\n{files['workload.yaml']}\n{bad}"
