@@ -183,3 +183,55 @@ def test_version_upgrade_sequence_parity(tmp_path):
     paths = {r.path for r in report}
     assert any("v1alpha1" in p for p in paths)
     assert any("v1alpha2" in p for p in paths)
+
+
+def test_custom_boilerplate_parity(tmp_path):
+    """A project initialized with a custom source-header license must
+    stay byte-identical too (the boilerplate flows through every
+    template's `{{ .Boilerplate }}`)."""
+    header = tmp_path / "header.txt"
+    header.write_text(
+        "// Copyright 2026 Fuzz Industries.\n// All rights reserved.\n"
+    )
+
+    workdir = tmp_path / "lic"
+    workdir.mkdir()
+    shutil.copytree(
+        os.path.join(FIXTURES, "standalone"), workdir / ".workloadConfig"
+    )
+    cwd = os.getcwd()
+    os.chdir(workdir)
+    try:
+        assert (
+            main(
+                [
+                    "init",
+                    "--workload-config",
+                    ".workloadConfig/workload.yaml",
+                    "--repo",
+                    "github.com/fuzz/lic",
+                    "--source-header-license",
+                    str(header),
+                ]
+            )
+            == 0
+        )
+        assert main(["create", "api"]) == 0
+
+        with open("apis/apps/v1alpha1/bookstore_types.go") as f:
+            assert f.read().startswith("// Copyright 2026 Fuzz Industries.")
+
+        project = Project.load(".")
+        processor = workload_config.parse(".workloadConfig/workload.yaml")
+        subcommand.create_api(processor)
+        ctx = _build_context(".", project, processor.workload)
+        report = diff_report(".", ctx, processor.workload)
+    finally:
+        os.chdir(cwd)
+
+    bad = {
+        r.path: ("MISSING" if r.missing else r.diff_lines)
+        for r in report
+        if r.missing or r.diff_lines != 0
+    }
+    assert not bad, bad
