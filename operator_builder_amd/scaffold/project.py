@@ -115,10 +115,17 @@ class Project:
                 raw["resources"].append(entry)
         raw["version"] = PROJECT_VERSION
 
+        dumper = getattr(yaml, "CSafeDumper", yaml.SafeDumper)
         with open(
             os.path.join(base_dir, PROJECT_FILE), "w", encoding="utf-8"
         ) as f:
-            yaml.safe_dump(raw, f, sort_keys=True, default_flow_style=False)
+            yaml.dump(
+                raw,
+                f,
+                Dumper=dumper,
+                sort_keys=True,
+                default_flow_style=False,
+            )
 
     # ---- resource registry ---------------------------------------------
 
