@@ -112,85 +112,6 @@ def _raw_string_lines(tokens: list[Token]) -> set[int]:
     return lines
 
 
-def _reindent(text: str) -> str:
-    """gofmt-style block re-indentation (tabs by bracket depth).
-
-    The raw template render loses/garbles indentation wherever control
-    actions owned the line ({{- .SourceCode }} splices, {{ end -}}
-    trims); gofmt recomputes indentation from syntax, so the oracle
-    must too.  Lines inside raw strings are untouched; `case`/`default`
-    dedent one level (gofmt switch style); continuation lines keep the
-    surrounding depth.
-    """
-    pass
-
-    try:
-        tokens = tokenize(text)
-    except GoLexError:
-        return text
-
-    protected: set[int] = set()
-    for t in tokens:
-        if t.kind in ("RAW_STRING", "COMMENT") and "\n" in t.text:
-            # interior + closing lines of raw strings and block comments
-            # keep their own indentation
-            for ln in range(t.line + 1, t.line + t.text.count("\n") + 1):
-                protected.add(ln)
-
-    by_line: dict[int, list] = {}
-    for t in tokens:
-        if t.kind == "OP" and t.text in "()[]{}":
-            by_line.setdefault(t.line, []).append(t.text)
-        elif t.kind not in ("NEWLINE",):
-            by_line.setdefault(t.line, []).append(None)
-
-    lines = text.split("\n")
-    out: list[str] = []
-    depth = 0
-    switch_depths: list[int] = []
-    for idx, raw in enumerate(lines):
-        lineno = idx + 1
-        s = raw.strip()
-        ops = by_line.get(lineno, [])
-
-        if lineno in protected or s == "":
-            out.append(raw if lineno in protected else "")
-            # still track depth from protected lines' brackets
-            for op in ops:
-                if op is None:
-                    continue
-                if op in "([{":
-                    depth += 1
-                else:
-                    depth = max(0, depth - 1)
-            continue
-
-        # leading run of closers dedents this line
-        leading_closers = 0
-        for op in ops:
-            if op is not None and op in ")]}":
-                leading_closers += 1
-            else:
-                break
-
-        indent = max(0, depth - leading_closers)
-        if s.startswith(("case ", "default:")) or s == "default:":
-            indent = max(0, indent - 1)
-        if s.startswith("switch ") or s.startswith("select {"):
-            switch_depths.append(depth)
-
-        out.append("\t" * indent + s)
-
-        for op in ops:
-            if op is None:
-                continue
-            if op in "([{":
-                depth += 1
-            else:
-                depth = max(0, depth - 1)
-
-    return "\n".join(out)
-
 
 @lru_cache(maxsize=512)
 def format_go(src: str) -> str:

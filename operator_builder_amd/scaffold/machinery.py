@@ -134,17 +134,6 @@ def _normalized_lines(text: str) -> list[str]:
     return [line.strip() for line in text.rstrip("\n").split("\n")]
 
 
-def _fragment_present(frag: str, file_norm: list[str]) -> bool:
-    """True iff the fragment's whitespace-normalized line sequence appears
-    as a contiguous run of lines in the file.
-
-    This is kubebuilder machinery's filterExistingValues semantics
-    (trimmed line-for-line equality) extended to multi-line fragments:
-    a fragment whose text merely appears as a *substring* of some longer
-    unrelated line (e.g. an import path that is a prefix of another)
-    must still be inserted.
-    """
-    return _fragment_present_norm(frag, file_norm)
 
 
 def insert_code_fragments(
@@ -203,6 +192,12 @@ def insert_code_fragments(
 
 
 def _fragment_present_norm(frag: str, file_norm: list[str]) -> bool:
+    """True iff the fragment's whitespace-normalized line sequence appears
+    as a contiguous run of lines in the file — kubebuilder machinery's
+    filterExistingValues semantics (trimmed line-for-line equality)
+    extended to multi-line fragments: a fragment whose text merely
+    appears as a *substring* of some longer unrelated line must still
+    be inserted."""
     needle = _normalized_lines(frag)
     if not any(needle):
         return False
