@@ -12,9 +12,12 @@ mostly-formatted text the templates render — has three visible effects:
   3. gofmt hygiene: no trailing whitespace, at most one consecutive
      blank line, exactly one trailing newline.
 
-This module reproduces those effects.  Full gofmt (re-indentation,
-alignment) is NOT attempted: the templates render already-gofmt-shaped
-text, and PARITY.md documents the residual risk.
+This module reproduces those effects, plus gofmt's structural
+normalizations (bracket-depth re-indentation, blank-line dropping
+before closing braces, forced blank between top-level declarations) in
+``_final_pass``.  Tabwriter column ALIGNMENT is not recomputed: the
+templates render already-aligned text; PARITY.md ("formatter model")
+documents the residual risk.
 """
 
 from __future__ import annotations
