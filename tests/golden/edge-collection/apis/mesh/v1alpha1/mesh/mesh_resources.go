@@ -115,3 +115,50 @@ func CreateDaemonSetEdgeSystemMeshAgent(
 
 	return resourceObjs, nil
 }
+
+// +kubebuilder:rbac:groups=apiextensions.k8s.io,resources=customresourcedefinitions,verbs=get;list;watch;create;update;patch;delete
+
+const CustomResourceDefinitionMeshpoliciesMeshExampleCom = "meshpolicies.mesh.example.com"
+
+// CreateCustomResourceDefinitionMeshpoliciesMeshExampleCom creates the meshpolicies.mesh.example.com CustomResourceDefinition resource.
+func CreateCustomResourceDefinitionMeshpoliciesMeshExampleCom(
+	parent *meshv1alpha1.ServiceMesh,
+	collection *edgeplatformv1alpha1.EdgePlatform,
+) ([]client.Object, error) {
+	resourceObjs := []client.Object{}
+	var resourceObj = &unstructured.Unstructured{
+		Object: map[string]interface{}{
+			"apiVersion": "apiextensions.k8s.io/v1",
+			"kind": "CustomResourceDefinition",
+			"metadata": map[string]interface{}{
+				"name": "meshpolicies.mesh.example.com",
+			},
+			"spec": map[string]interface{}{
+				"group": "mesh.example.com",
+				"names": map[string]interface{}{
+					"kind": "MeshPolicy",
+					"listKind": "MeshPolicyList",
+					"plural": "meshpolicies",
+					"singular": "meshpolicy",
+				},
+				"scope": "Namespaced",
+				"versions": []interface{}{
+					map[string]interface{}{
+						"name": "v1",
+						"served": true,
+						"storage": true,
+						"schema": map[string]interface{}{
+							"openAPIV3Schema": map[string]interface{}{
+								"type": "object",
+							},
+						},
+					},
+				},
+			},
+		},
+	}
+
+	resourceObjs = append(resourceObjs, resourceObj)
+
+	return resourceObjs, nil
+}

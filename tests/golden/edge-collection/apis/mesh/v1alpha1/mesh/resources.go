@@ -114,6 +114,7 @@ var CreateFuncs = []func(
 ) ([]client.Object, error) {
 	CreateClusterRoleMeshAgent,
 	CreateDaemonSetEdgeSystemMeshAgent,
+	CreateCustomResourceDefinitionMeshpoliciesMeshExampleCom,
 }
 
 // InitFuncs is an array of functions that are called prior to starting the controller manager.  This is
@@ -128,6 +129,7 @@ var InitFuncs = []func(
 	*meshv1alpha1.ServiceMesh,
 	*edgeplatformv1alpha1.EdgePlatform,
 ) ([]client.Object, error) {
+	CreateCustomResourceDefinitionMeshpoliciesMeshExampleCom,
 }
 
 func ConvertWorkload(component, collection workload.Workload) (
