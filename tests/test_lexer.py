@@ -21,6 +21,71 @@ def lex_all(src):
 
 
 CASES = {
+    "marker arg with two scopes": (
+        "+galaxy:planet:name=earth",
+        [
+            (T.MARKER_START, "+"),
+            (T.SCOPE, "galaxy"),
+            (T.SEPARATOR, ":"),
+            (T.SCOPE, "planet"),
+            (T.SEPARATOR, ":"),
+            (T.ARG, "name"),
+            (T.ARG_ASSIGNMENT, "="),
+            (T.STRING_LITERAL, "earth"),
+            (T.MARKER_END, "\n"),
+            (T.EOF, ""),
+        ],
+    ),
+    "marker in yaml comment with white space": (
+        "#     +hello:world",
+        [
+            (T.COMMENT, "#"),
+            (T.MARKER_START, "+"),
+            (T.SCOPE, "hello"),
+            (T.SEPARATOR, ":"),
+            (T.ARG, "world"),
+            (T.SYNTHETIC_BOOL_LITERAL, "true"),
+            (T.MARKER_END, "\n"),
+            (T.EOF, ""),
+        ],
+    ),
+    "marker with two scopes and two args": (
+        "+galaxy:planet:name=earth,solar-system=milky-way",
+        [
+            (T.MARKER_START, "+"),
+            (T.SCOPE, "galaxy"),
+            (T.SEPARATOR, ":"),
+            (T.SCOPE, "planet"),
+            (T.SEPARATOR, ":"),
+            (T.ARG, "name"),
+            (T.ARG_ASSIGNMENT, "="),
+            (T.STRING_LITERAL, "earth"),
+            (T.ARG_DELIMITER, ","),
+            (T.ARG, "solar-system"),
+            (T.ARG_ASSIGNMENT, "="),
+            (T.STRING_LITERAL, "milky-way"),
+            (T.MARKER_END, "\n"),
+            (T.EOF, ""),
+        ],
+    ),
+    "marker with two scopes and two args one of which is a flag": (
+        "+galaxy:planet:name=earth,current-location",
+        [
+            (T.MARKER_START, "+"),
+            (T.SCOPE, "galaxy"),
+            (T.SEPARATOR, ":"),
+            (T.SCOPE, "planet"),
+            (T.SEPARATOR, ":"),
+            (T.ARG, "name"),
+            (T.ARG_ASSIGNMENT, "="),
+            (T.STRING_LITERAL, "earth"),
+            (T.ARG_DELIMITER, ","),
+            (T.ARG, "current-location"),
+            (T.SYNTHETIC_BOOL_LITERAL, "true"),
+            (T.MARKER_END, "\n"),
+            (T.EOF, ""),
+        ],
+    ),
     "marker start": (
         "+test:flag",
         [
