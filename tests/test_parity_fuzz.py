@@ -323,3 +323,116 @@ def test_random_collections_stay_byte_identical(tmp_path_factory, files):
         if r.missing or r.diff_lines != 0
     }
     assert not bad, f"diverged for:\n{files['workload.yaml']}\n{bad}"
+
+
+@st.composite
+def edge_setups(draw):
+    """Standalone workloads exercising the semantic core: dotted nested
+    field paths, replace= regex splices, multi-document manifests,
+    descriptions, and resource markers on field markers."""
+    group = draw(kinds_names)
+    kind = draw(kinds_names).capitalize() + "Edge"
+
+    parts = draw(st.lists(names, min_size=2, max_size=4, unique=True))
+    dotted = ".".join(parts)
+    replace_word = draw(names)
+    value = f"{replace_word}-suffix"
+    flag_field = draw(names.filter(lambda n: n not in parts))
+
+    use_resource_marker = draw(st.booleans())
+    rm = ""
+    if use_resource_marker:
+        rm = (
+            f"# +operator-builder:resource:field={flag_field},"
+            f"value=true,include={'true' if draw(st.booleans()) else 'false'}\n"
+        )
+
+    config = (
+        f"name: {draw(names)}-edge\n"
+        "kind: StandaloneWorkload\n"
+        "spec:\n"
+        "  api:\n"
+        "    domain: example.com\n"
+        f"    group: {group}\n"
+        "    version: v1alpha1\n"
+        f"    kind: {kind}\n"
+        "    clusterScoped: false\n"
+        "  resources:\n"
+        "  - r.yaml\n"
+    )
+
+    manifest = (
+        f"{rm}"
+        "apiVersion: v1\n"
+        "kind: ConfigMap\n"
+        "metadata:\n"
+        "  name: edge-config\n"
+        "  namespace: default\n"
+        "data:\n"
+        f"  # +operator-builder:field:name={dotted},type=string,"
+        f'replace="{replace_word}",default="{value}",'
+        'description="a nested replaced value"\n'
+        f'  nested: "{value}"\n'
+        f"  # +operator-builder:field:name={flag_field},type=bool,"
+        "default=true\n"
+        "  flag: true\n"
+        "---\n"
+        "apiVersion: v1\n"
+        "kind: Secret\n"
+        "metadata:\n"
+        "  name: edge-secret\n"
+        "  namespace: default\n"
+        "type: Opaque\n"
+        "stringData:\n"
+        '  token: "abc"\n'
+    )
+    return config, manifest
+
+
+@settings(
+    max_examples=10,
+    deadline=None,
+    derandomize=True,
+    suppress_health_check=[HealthCheck.too_slow],
+)
+@given(edge_setups())
+def test_edge_semantics_stay_byte_identical(tmp_path_factory, setup):
+    config, manifest = setup
+    workdir = tmp_path_factory.mktemp("parityfuzzedge")
+    cfg_dir = workdir / ".workloadConfig"
+    cfg_dir.mkdir()
+    (cfg_dir / "workload.yaml").write_text(config)
+    (cfg_dir / "r.yaml").write_text(manifest)
+
+    cwd = os.getcwd()
+    os.chdir(workdir)
+    try:
+        assert (
+            main(
+                [
+                    "init",
+                    "--workload-config",
+                    ".workloadConfig/workload.yaml",
+                    "--repo",
+                    "github.com/fuzz/edge",
+                ]
+            )
+            == 0
+        ), config + manifest
+        assert main(["create", "api"]) == 0, config + manifest
+
+        project = Project.load(".")
+        processor = workload_config.parse(".workloadConfig/workload.yaml")
+        subcommand.create_api(processor)
+        ctx = _build_context(".", project, processor.workload)
+        report = diff_report(".", ctx, processor.workload)
+    finally:
+        os.chdir(cwd)
+        shutil.rmtree(workdir, ignore_errors=True)
+
+    bad = {
+        r.path: ("MISSING" if r.missing else r.diff_lines)
+        for r in report
+        if r.missing or r.diff_lines != 0
+    }
+    assert not bad, f"diverged for:\n{config}\n{manifest}\n{bad}"
