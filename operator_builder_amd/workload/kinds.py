@@ -77,6 +77,18 @@ WORKLOAD_KINDS = (
 )
 
 
+def _scalar_str(value) -> str:
+    """yaml.v3 semantics for decoding a YAML scalar into a Go string
+    field: any scalar unmarshals to its canonical string form (`false`
+    -> "false", `123` -> "123"); absent/null -> "".  PyYAML resolves
+    scalars to native types first, so coerce back."""
+    if value is None:
+        return ""
+    if isinstance(value, bool):
+        return "true" if value else "false"
+    return str(value)
+
+
 @dataclass
 class WorkloadAPISpec:
     """Shared `spec.api` block (reference workload.go:81-88)."""
@@ -95,10 +107,10 @@ class WorkloadAPISpec:
             "spec.api",
         )
         return cls(
-            domain=raw.get("domain", "") or "",
-            group=raw.get("group", "") or "",
-            version=raw.get("version", "") or "",
-            kind=raw.get("kind", "") or "",
+            domain=_scalar_str(raw.get("domain")),
+            group=_scalar_str(raw.get("group")),
+            version=_scalar_str(raw.get("version")),
+            kind=_scalar_str(raw.get("kind")),
             cluster_scoped=bool(raw.get("clusterScoped", False)),
         )
 
@@ -132,8 +144,8 @@ def _companion_from_dict(raw: Optional[dict], where: str) -> CLI:
         return CLI()
     _check_fields(raw, {"name", "description"}, where)
     return CLI(
-        name=raw.get("name", "") or "",
-        description=raw.get("description", "") or "",
+        name=_scalar_str(raw.get("name")),
+        description=_scalar_str(raw.get("description")),
     )
 
 
@@ -519,7 +531,7 @@ class StandaloneWorkload(Workload):
             f"{cls.kind}.spec",
         )
         return cls(
-            name=raw.get("name", "") or "",
+            name=_scalar_str(raw.get("name")),
             api=WorkloadAPISpec.from_dict(spec_raw.get("api") or {}),
             companion_cli_rootcmd=_companion_from_dict(
                 spec_raw.get("companionCliRootcmd"),
@@ -594,7 +606,7 @@ class ComponentWorkload(Workload):
             f"{cls.kind}.spec",
         )
         return cls(
-            name=raw.get("name", "") or "",
+            name=_scalar_str(raw.get("name")),
             api=WorkloadAPISpec.from_dict(spec_raw.get("api") or {}),
             companion_cli_subcmd=_companion_from_dict(
                 spec_raw.get("companionCliSubcmd"),
@@ -699,7 +711,7 @@ class WorkloadCollection(Workload):
             f"{cls.kind}.spec",
         )
         return cls(
-            name=raw.get("name", "") or "",
+            name=_scalar_str(raw.get("name")),
             api=WorkloadAPISpec.from_dict(spec_raw.get("api") or {}),
             companion_cli_rootcmd=_companion_from_dict(
                 spec_raw.get("companionCliRootcmd"),

@@ -20,6 +20,8 @@ sys.path.insert(
 
 from hypothesis import HealthCheck, given, settings  # noqa: E402
 
+from operator_builder_amd.parity.oracle import reference_available  # noqa: E402
+
 import test_parity_fuzz as m  # noqa: E402
 
 
@@ -36,6 +38,9 @@ class _Factory:
 
 
 def main() -> int:
+    if not reference_available():
+        print("skip: reference checkout not available")
+        return 0
     budget = int(sys.argv[1]) if len(sys.argv) > 1 else 100
     suites = [
         ("standalone", m.workload_setups,
