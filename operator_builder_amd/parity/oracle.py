@@ -565,20 +565,20 @@ def _render_cli_sub(
         init_name, init_descr = sub_cmd_obj.name, sub_cmd_obj.description
     p = sub_path("init")
     if p not in files:  # SkipFile (matches this repo's cmd_init_sub)
-        files[p] = _tpl(
-        "cli/cmd_init_sub.go",
-        "cmdInitSub",
-        (cli_tpl.OB_IMPORTS_MARKER, cli_tpl.OB_VERSIONMAP_MARKER),
-    ).render(
-        {
-            **base,
-            "Builder": builder,
-            "RootCmd": root_cmd,
-            "SubCmd": sub_cmd,
-            "InitCommandName": init_name,
-            "InitCommandDescr": init_descr,
-        }
-    )
+            files[p] = _tpl(
+            "cli/cmd_init_sub.go",
+            "cmdInitSub",
+            (cli_tpl.OB_IMPORTS_MARKER, cli_tpl.OB_VERSIONMAP_MARKER),
+        ).render(
+            {
+                **base,
+                "Builder": builder,
+                "RootCmd": root_cmd,
+                "SubCmd": sub_cmd,
+                "InitCommandName": init_name,
+                "InitCommandDescr": init_descr,
+            }
+        )
     files[p] = insert_code_fragments(
         files[p],
         {
@@ -614,27 +614,27 @@ def _render_cli_sub(
         col = workload.get_collection()
         p = sub_path("generate")
         if p not in files:  # SkipFile
-            files[p] = _tpl(
-            "cli/cmd_generate_sub.go",
-            "cmdGenerateSub",
-            (
-                cli_tpl.OB_IMPORTS_MARKER,
-                cli_tpl.OB_VERSIONMAP_MARKER,
-            ),
-        ).render(
-            {
-                **base,
-                "Builder": builder,
-                "RootCmd": root_cmd,
-                "SubCmd": sub_cmd,
-                "Collection": builder_shape(col) if col else None,
-                "UseCollectionManifestFlag": use_collection,
-                "UseWorkloadManifestFlag": use_workload,
-                "GenerateCommandName": gen_name,
-                "GenerateCommandDescr": gen_descr,
-                "GenerateFuncInputs": gen_inputs,
-            }
-        )
+                files[p] = _tpl(
+                "cli/cmd_generate_sub.go",
+                "cmdGenerateSub",
+                (
+                    cli_tpl.OB_IMPORTS_MARKER,
+                    cli_tpl.OB_VERSIONMAP_MARKER,
+                ),
+            ).render(
+                {
+                    **base,
+                    "Builder": builder,
+                    "RootCmd": root_cmd,
+                    "SubCmd": sub_cmd,
+                    "Collection": builder_shape(col) if col else None,
+                    "UseCollectionManifestFlag": use_collection,
+                    "UseWorkloadManifestFlag": use_workload,
+                    "GenerateCommandName": gen_name,
+                    "GenerateCommandDescr": gen_descr,
+                    "GenerateFuncInputs": gen_inputs,
+                }
+            )
         files[p] = insert_code_fragments(
             files[p],
             {
@@ -654,16 +654,16 @@ def _render_cli_sub(
         ver_name, ver_descr = sub_cmd_obj.name, sub_cmd_obj.description
     p = sub_path("version")
     if p not in files:  # SkipFile
-        files[p] = _tpl("cli/cmd_version_sub.go", "cmdVersionSub").render(
-        {
-            **base,
-            "Builder": builder,
-            "RootCmd": root_cmd,
-            "SubCmd": sub_cmd,
-            "VersionCommandName": ver_name,
-            "VersionCommandDescr": ver_descr,
-        }
-    )
+            files[p] = _tpl("cli/cmd_version_sub.go", "cmdVersionSub").render(
+            {
+                **base,
+                "Builder": builder,
+                "RootCmd": root_cmd,
+                "SubCmd": sub_cmd,
+                "VersionCommandName": ver_name,
+                "VersionCommandDescr": ver_descr,
+            }
+        )
     files[p] = insert_code_fragments(
         files[p],
         {cli_tpl.OB_APIVERSIONS_MARKER: [f'"{version}",\n']},
