@@ -19,4 +19,4 @@ Package map (reference parity):
   cli/       -> cmd/operator-builder + pkg/cli
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"

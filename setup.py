@@ -6,7 +6,7 @@ from setuptools import find_packages, setup
 
 setup(
     name="operator-builder-amd",
-    version="0.1.0",
+    version="0.2.0",
     description=(
         "Kubernetes operator code generator (operator-builder capability "
         "surface, rebuilt from scratch)"
