@@ -140,7 +140,9 @@ spec:
 def test_version_flag():
     result = run_cli("--version")
     assert result.returncode == 0
-    assert "0.1.0" in result.stdout
+    from operator_builder_amd import __version__
+
+    assert __version__ in result.stdout
 
 
 def test_bare_invocation_prints_help():
