@@ -152,24 +152,24 @@ def test_version_upgrade_sequence_parity(tmp_path):
         subcommand.create_api(processor_v1)
         ctx_v1 = _build_context(".", project, processor_v1.workload)
 
-        # bump the API version and regenerate (documented workflow)
-        cfg.write_text(
-            cfg.read_text().replace("version: v1alpha1", "version: v1alpha2")
-        )
-        assert main(["create", "api", "--force"]) == 0
+        # bump the API version twice and regenerate each time
+        # (documented workflow, run repeatedly)
+        runs = [(ctx_v1, processor_v1.workload)]
+        for old_v, new_v in (("v1alpha1", "v1alpha2"), ("v1alpha2", "v1alpha3")):
+            cfg.write_text(
+                cfg.read_text().replace(f"version: {old_v}", f"version: {new_v}")
+            )
+            assert main(["create", "api", "--force"]) == 0
 
-        project = Project.load(".")
-        processor_v2 = workload_config.parse(str(cfg))
-        subcommand.create_api(processor_v2)
-        ctx_v2 = _build_context(".", project, processor_v2.workload)
+            project = Project.load(".")
+            processor_n = workload_config.parse(str(cfg))
+            subcommand.create_api(processor_n)
+            runs.append(
+                (_build_context(".", project, processor_n.workload),
+                 processor_n.workload)
+            )
 
-        report = diff_report_sequence(
-            ".",
-            [
-                (ctx_v1, processor_v1.workload),
-                (ctx_v2, processor_v2.workload),
-            ],
-        )
+        report = diff_report_sequence(".", runs)
     finally:
         os.chdir(cwd)
 
@@ -181,8 +181,8 @@ def test_version_upgrade_sequence_parity(tmp_path):
     assert not bad, bad
     # both versions' files must be part of the oracle-covered set
     paths = {r.path for r in report}
-    assert any("v1alpha1" in p for p in paths)
-    assert any("v1alpha2" in p for p in paths)
+    for v in ("v1alpha1", "v1alpha2", "v1alpha3"):
+        assert any(v in p for p in paths), v
 
 
 def test_custom_boilerplate_parity(tmp_path):
