@@ -3,12 +3,16 @@
 
 PYTHON ?= python3
 
-.PHONY: all test test-unit test-func test-gpu bench smoke lint parity release clean
+.PHONY: all test test-fast test-unit test-func test-gpu bench smoke lint parity release clean
 
 all: lint test
 
 test:
 	$(PYTHON) -m pytest tests/ -q -m "not gpu"
+
+# parallel run (pytest-xdist); the suite is worker-safe
+test-fast:
+	$(PYTHON) -m pytest tests/ -q -m "not gpu" -n auto
 
 # unit tiers only (marker engine, yamlast, domain model)
 test-unit:
