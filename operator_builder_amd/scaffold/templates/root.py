@@ -1,12 +1,15 @@
 """Project-root templates: main.go, go.mod, Makefile, Dockerfile, README.
 
 Parity targets: reference templates/main.go:22-282 (Main + MainUpdater),
-with one correctness deviation: the reference's mainTemplate imports
-sigs.k8s.io/controller-runtime/pkg/controller without using it (an
-unused import is a Go compile error), so it is omitted here.
-Other parity targets:
 templates/gomod.go:23-66, templates/makefile.go, templates/dockerfile.go,
 templates/readme.go.
+
+Note on the reference mainTemplate's unused
+sigs.k8s.io/controller-runtime/pkg/controller import (main.go:178):
+machinery's imports.Process (goimports) removes unused imports at
+scaffold time, so the reference's real output never contains it; this
+module omits it to the same effect, and the scaffold pipeline's own
+format_go would strip it anyway (PARITY.md "resolved with evidence").
 """
 
 from __future__ import annotations
