@@ -239,6 +239,23 @@ def cmd_validate(args) -> int:
     return 0
 
 
+def cmd_check(args) -> int:
+    """Static compile gate over a generated tree (extension beyond the
+    reference's commands): token-level Go checks — delimiter balance,
+    package clauses, declared-vs-used imports, missing stdlib
+    qualifiers, duplicate top-level funcs, per-directory package
+    consistency — the offline stand-in for `go build ./...`."""
+    from ..golang.check import check_tree
+
+    issues = check_tree(args.directory)
+    for issue in issues:
+        print(issue)
+    if issues:
+        raise CLIError(f"{len(issues)} issue(s) found in {args.directory}")
+    print(f"ok: no issues found in {args.directory}")
+    return 0
+
+
 def cmd_version(args) -> int:
     print(f"operator-builder-amd version {__version__}")
     return 0
@@ -376,6 +393,14 @@ def build_parser() -> argparse.ArgumentParser:
     )
     p_validate.add_argument("--workload-config", required=True)
     p_validate.set_defaults(func=cmd_validate)
+
+    # check (extension)
+    p_check = sub.add_parser(
+        "check",
+        help="Run the static Go compile gate over a generated tree",
+    )
+    p_check.add_argument("--directory", default=".")
+    p_check.set_defaults(func=cmd_check)
 
     # version
     p_version = sub.add_parser("version", help="Print version information")

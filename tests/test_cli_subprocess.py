@@ -149,3 +149,40 @@ def test_bare_invocation_prints_help():
     result = run_cli()
     assert result.returncode == 0
     assert "usage: operator-builder" in result.stdout
+
+
+def test_check_command(tmp_path):
+    """`operator-builder check` (extension) runs the static Go gate."""
+    import shutil
+
+    FIXTURES = os.path.join(REPO, "tests", "fixtures")
+
+    workdir = tmp_path / "chk"
+    workdir.mkdir()
+    shutil.copytree(
+        os.path.join(FIXTURES, "standalone"), workdir / ".workloadConfig"
+    )
+    assert (
+        run_cli(
+            "init",
+            "--workload-config",
+            ".workloadConfig/workload.yaml",
+            "--repo",
+            "github.com/x/chk",
+            cwd=workdir,
+        ).returncode
+        == 0
+    )
+    assert run_cli("create", "api", cwd=workdir).returncode == 0
+
+    result = run_cli("check", cwd=workdir)
+    assert result.returncode == 0, result.stderr
+    assert "ok: no issues found" in result.stdout
+
+    # break a file and expect a nonzero exit with the issue printed
+    broken = next((workdir / "apis").rglob("*_types.go"))
+    broken.write_text(broken.read_text() + "\nfunc oops() {\n")
+    result = run_cli("check", cwd=workdir)
+    assert result.returncode == 1
+    assert "unclosed delimiter" in result.stdout
+    assert "FATAL" in result.stderr
