@@ -254,3 +254,41 @@ spec:
         assert processor.children[0].workload.config_path.endswith(
             "comp.yaml"
         )
+
+
+def test_null_scalar_in_required_string_field_rejected(tmp_path):
+    """`group: null` decodes to the empty string (yaml.v3 semantics) and
+    fails required-field validation — same outcome as the reference."""
+    import pytest as _pytest
+
+    from operator_builder_amd.workload import config as workload_config
+
+    cfg = tmp_path / "w.yaml"
+    cfg.write_text(
+        "name: x\nkind: StandaloneWorkload\nspec:\n  api:\n"
+        "    domain: example.com\n    group: null\n    version: v1\n"
+        "    kind: App\n  resources:\n  - r.yaml\n"
+    )
+    (tmp_path / "r.yaml").write_text(
+        "apiVersion: v1\nkind: ConfigMap\nmetadata:\n  name: c\n"
+    )
+    with _pytest.raises(Exception, match="spec.api.group"):
+        workload_config.parse(str(cfg))
+
+
+def test_boolish_scalar_in_string_field_coerced(tmp_path):
+    """`group: false` decodes to the string "false" (yaml.v3 decodes any
+    scalar into a Go string field as its canonical text)."""
+    from operator_builder_amd.workload import config as workload_config
+
+    cfg = tmp_path / "w.yaml"
+    cfg.write_text(
+        "name: x\nkind: StandaloneWorkload\nspec:\n  api:\n"
+        "    domain: example.com\n    group: false\n    version: v1\n"
+        "    kind: App\n  resources:\n  - r.yaml\n"
+    )
+    (tmp_path / "r.yaml").write_text(
+        "apiVersion: v1\nkind: ConfigMap\nmetadata:\n  name: c\n"
+    )
+    processor = workload_config.parse(str(cfg))
+    assert processor.workload.get_api_group() == "false"
