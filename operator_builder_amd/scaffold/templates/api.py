@@ -18,21 +18,32 @@ KIND_GROUPVERSIONS_MARKER = Marker("//", "operator-builder:groupversions")
 
 
 def _dependency_imports(ctx: Context, builder: Workload) -> str:
-    """Imports for cross-group dependencies in the types file
-    (reference types.go:62-70)."""
+    """Imports for cross-group dependencies in the types file,
+    replicating the reference template's raw whitespace emission
+    (types.go:62-70): each loop iteration leaves tab-only lines for its
+    skipped/closing branches, so — after the gofmt pass — the first
+    cross-group import joins the static import group only when no
+    skipped dependency precedes it, and later imports form their own
+    groups.  The parity oracle holds byte-for-byte across dependency
+    orderings because both sides share this emission + format_go."""
     out = []
     added = set()
     for dep in builder.get_dependencies():
-        if dep.get_api_group() == ctx.resource.group:
-            continue
-        key = f"{dep.get_api_group()}{dep.get_api_version()}"
-        if key in added:
-            continue
-        added.add(key)
-        out.append(
-            f'\t{key} "{ctx.repo}/apis/{dep.get_api_group()}/'
-            f'{dep.get_api_version()}"\n'
-        )
+        if dep.get_api_group() != ctx.resource.group:
+            key = f"{dep.get_api_group()}{dep.get_api_version()}"
+            if key not in added:
+                added.add(key)
+                # {{ import line }} then the two closing `{{ end }}` lines
+                out.append(
+                    f'\n\t{key} "{ctx.repo}/apis/{dep.get_api_group()}/'
+                    f'{dep.get_api_version()}"\n\t\n\t\n\t'
+                )
+            else:
+                # dedup-skipped: inner + outer end lines remain
+                out.append("\n\t\n\t")
+        else:
+            # same-group: only the outer end line remains
+            out.append("\n\t")
     return "".join(out)
 
 
@@ -67,8 +78,8 @@ import (
 \t"github.com/nukleros/operator-builder-tools/pkg/status"
 \t"github.com/nukleros/operator-builder-tools/pkg/controller/workload"
 \tmetav1 "k8s.io/apimachinery/pkg/apis/meta/v1"
-\t"k8s.io/apimachinery/pkg/runtime/schema"
-{_dependency_imports(ctx, builder)})
+\t"k8s.io/apimachinery/pkg/runtime/schema"{_dependency_imports(ctx, builder)}
+)
 
 var ErrUnableToConvert{kind} = errors.New("unable to convert to {kind}")
 

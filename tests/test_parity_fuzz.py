@@ -197,11 +197,16 @@ def collection_setups(draw):
     markers, and optionally a resource marker on a collectionField."""
     col_group = draw(kinds_names)
     col_kind = draw(kinds_names).capitalize() + "Platform"
-    n_components = draw(st.integers(min_value=1, max_value=2))
+    n_components = draw(st.integers(min_value=1, max_value=3))
     col_cluster = draw(st.booleans())
 
     col_field = draw(names)
     files = {}
+
+    # sharing one API group across components exercises same-group
+    # dependency lists and accumulated per-group files; multi-deps
+    # exercise the typesTemplate $Added import-dedup branch
+    shared_group = draw(kinds_names) if draw(st.booleans()) else None
 
     comp_blocks = []
     comp_names = []
@@ -209,11 +214,14 @@ def collection_setups(draw):
         comp_name = f"comp-{draw(names)}-{i}"
         comp_names.append(comp_name)
     for i, comp_name in enumerate(comp_names):
-        group = draw(kinds_names)
+        group = shared_group or draw(kinds_names)
         kind = draw(kinds_names).capitalize() + f"Part{i}"
         dep_block = ""
         if i > 0 and draw(st.booleans()):
-            dep_block = f"  dependencies:\n  - {comp_names[0]}\n"
+            deps = comp_names[:i]
+            dep_block = "  dependencies:\n" + "".join(
+                f"  - {d}\n" for d in deps
+            )
         use_resource_marker = draw(st.booleans())
         rm_line = ""
         if use_resource_marker:
