@@ -215,6 +215,7 @@ def collection_setups(draw):
         comp_names.append(comp_name)
     for i, comp_name in enumerate(comp_names):
         group = shared_group or draw(kinds_names)
+        version = "v1alpha" + str(draw(st.integers(min_value=1, max_value=2)))
         kind = draw(kinds_names).capitalize() + f"Part{i}"
         dep_block = ""
         if i > 0 and draw(st.booleans()):
@@ -236,7 +237,7 @@ def collection_setups(draw):
             "spec:\n"
             "  api:\n"
             f"    group: {group}\n"
-            "    version: v1alpha1\n"
+            f"    version: {version}\n"
             f"    kind: {kind}\n"
             f"    clusterScoped: {'true' if draw(st.booleans()) else 'false'}\n"
             "  companionCliSubcmd:\n"
