@@ -208,3 +208,62 @@ def test_collection_itself_has_no_collection_ref(project):
         project, "config/samples/platforms_v1alpha1_cloudplatform.yaml"
     )
     assert "collection" not in sample
+
+
+def test_mixed_group_dependency_import_grouping(tmp_path, monkeypatch):
+    """A component whose dependency list holds a same-group dep BEFORE a
+    cross-group dep renders the cross-group import in its own import
+    group (the reference template's skipped-iteration whitespace becomes
+    a group separator after gofmt; found by the collection parity fuzz)."""
+    cfg = tmp_path / ".workloadConfig"
+    for sub in ("a", "b", "c"):
+        (cfg / sub).mkdir(parents=True)
+    (cfg / "workload.yaml").write_text(
+        "name: mix\nkind: WorkloadCollection\nspec:\n  api:\n"
+        "    domain: example.com\n    group: root\n    version: v1\n"
+        "    kind: Mix\n    clusterScoped: false\n"
+        "  resources: []\n"
+        "  componentFiles:\n  - a/c.yaml\n  - b/c.yaml\n  - c/c.yaml\n"
+    )
+    manifest = (
+        "apiVersion: v1\nkind: ConfigMap\nmetadata:\n  name: {n}\n"
+        "  namespace: default\ndata:\n"
+        '  # +operator-builder:field:name=f{n},type=string,default="x"\n'
+        '  k: "x"\n'
+    )
+    for name, group, deps in (
+        ("aaa", "grp", ""),
+        ("bbb", "other", ""),
+        ("ccc", "grp", "  dependencies:\n  - aaa\n  - bbb\n"),
+    ):
+        sub = {"aaa": "a", "bbb": "b", "ccc": "c"}[name]
+        (cfg / sub / "c.yaml").write_text(
+            f"name: {name}\nkind: ComponentWorkload\nspec:\n  api:\n"
+            f"    group: {group}\n    version: v1\n"
+            f"    kind: {name.capitalize()}K\n    clusterScoped: false\n"
+            f"{deps}"
+            "  resources:\n  - r.yaml\n"
+        )
+        (cfg / sub / "r.yaml").write_text(manifest.format(n=name))
+
+    monkeypatch.chdir(tmp_path)
+    assert (
+        main(
+            [
+                "init",
+                "--workload-config",
+                ".workloadConfig/workload.yaml",
+                "--repo",
+                "github.com/x/mix",
+            ]
+        )
+        == 0
+    )
+    assert main(["create", "api"]) == 0
+
+    types = read(str(tmp_path), "apis/grp/v1/ccck_types.go")
+    # cross-group import separated from the static group by a blank line
+    assert (
+        '"k8s.io/apimachinery/pkg/runtime/schema"\n\n'
+        '\totherv1 "github.com/x/mix/apis/other/v1"\n)' in types
+    )
