@@ -49,6 +49,8 @@ def main() -> int:
          m.test_random_collections_stay_byte_identical),
         ("edge-semantics", m.edge_setups,
          m.test_edge_semantics_stay_byte_identical),
+        ("upgrade-sequences", m.upgrade_plans,
+         m.test_random_upgrade_sequences_stay_byte_identical),
     ]
     for name, strategy, test in suites:
         inner = test.hypothesis.inner_test

@@ -453,3 +453,127 @@ def test_edge_semantics_stay_byte_identical(tmp_path_factory, setup):
         if r.missing or r.diff_lines != 0
     }
     assert not bad, f"diverged for:\n{config}\n{manifest}\n{bad}"
+
+
+@st.composite
+def upgrade_plans(draw):
+    """An initial standalone workload plus 1-2 evolution steps: version
+    bumps and/or manifest field additions, each followed by
+    `create api --force` (the documented update workflows)."""
+    group = draw(kinds_names)
+    kind = draw(kinds_names).capitalize() + "Up"
+    base_field = draw(names)
+    steps = []
+    n_steps = draw(st.integers(min_value=1, max_value=2))
+    version_idx = 1
+    for _ in range(n_steps):
+        action = draw(st.sampled_from(["bump", "add_field", "both"]))
+        new_field = draw(names.filter(lambda n, b=base_field: n != b))
+        if action in ("bump", "both"):
+            version_idx += 1
+        steps.append((action, new_field, version_idx))
+    return group, kind, base_field, steps
+
+
+@settings(
+    max_examples=8,
+    deadline=None,
+    derandomize=True,
+    suppress_health_check=[HealthCheck.too_slow],
+)
+@given(upgrade_plans())
+def test_random_upgrade_sequences_stay_byte_identical(
+    tmp_path_factory, plan
+):
+    from operator_builder_amd.parity.oracle import diff_report_sequence
+
+    group, kind, base_field, steps = plan
+    workdir = tmp_path_factory.mktemp("parityfuzzup")
+    cfg_dir = workdir / ".workloadConfig"
+    cfg_dir.mkdir()
+
+    def write_config(version):
+        (cfg_dir / "workload.yaml").write_text(
+            f"name: {group}-up\n"
+            "kind: StandaloneWorkload\n"
+            "spec:\n"
+            "  api:\n"
+            "    domain: example.com\n"
+            f"    group: {group}\n"
+            f"    version: v1alpha{version}\n"
+            f"    kind: {kind}\n"
+            "    clusterScoped: false\n"
+            "  resources:\n"
+            "  - r.yaml\n"
+        )
+
+    def write_manifest(fields):
+        lines = [
+            "apiVersion: v1",
+            "kind: ConfigMap",
+            "metadata:",
+            "  name: up-config",
+            "  namespace: default",
+            "data:",
+        ]
+        for f in fields:
+            lines.append(
+                f"  # +operator-builder:field:name={f},type=string,"
+                'default="x"'
+            )
+            lines.append(f'  {f}: "x"')
+        (cfg_dir / "r.yaml").write_text("\n".join(lines) + "\n")
+
+    cwd = os.getcwd()
+    os.chdir(workdir)
+    try:
+        write_config(1)
+        fields = [base_field]
+        write_manifest(fields)
+        assert (
+            main(
+                [
+                    "init",
+                    "--workload-config",
+                    ".workloadConfig/workload.yaml",
+                    "--repo",
+                    "github.com/fuzz/up",
+                ]
+            )
+            == 0
+        )
+        assert main(["create", "api"]) == 0
+
+        def snapshot():
+            project = Project.load(".")
+            processor = workload_config.parse(
+                ".workloadConfig/workload.yaml"
+            )
+            subcommand.create_api(processor)
+            return (
+                _build_context(".", project, processor.workload),
+                processor.workload,
+            )
+
+        runs = [snapshot()]
+        for action, new_field, version_idx in steps:
+            if action in ("bump", "both"):
+                write_config(version_idx)
+            if action in ("add_field", "both"):
+                if new_field not in fields:
+                    fields.append(new_field)
+                write_manifest(fields)
+            assert main(["create", "api", "--force"]) == 0, plan
+            runs.append(snapshot())
+
+        report = diff_report_sequence(".", runs)
+    finally:
+        os.chdir(cwd)
+        shutil.rmtree(workdir, ignore_errors=True)
+
+    bad = {
+        r.path: ("MISSING" if r.missing else r.diff_lines)
+        for r in report
+        if r.missing or r.diff_lines != 0
+    }
+    assert not bad, f"diverged for plan {plan}:\n{bad}"
