@@ -261,6 +261,14 @@ def collection_setups(draw):
         )
         comp_blocks.append(f"  - {comp_name}/component.yaml")
 
+    # collections may omit the companion CLI entirely (no cmd/ tree)
+    cli_block = ""
+    if draw(st.booleans()):
+        cli_block = (
+            "  companionCliRootcmd:\n"
+            f"    name: {draw(names)}ctl\n"
+            "    description: Manage the platform\n"
+        )
     files["workload.yaml"] = (
         f"name: {draw(names)}-collection\n"
         "kind: WorkloadCollection\n"
@@ -271,9 +279,7 @@ def collection_setups(draw):
         "    version: v1alpha1\n"
         f"    kind: {col_kind}\n"
         f"    clusterScoped: {'true' if col_cluster else 'false'}\n"
-        "  companionCliRootcmd:\n"
-        f"    name: {draw(names)}ctl\n"
-        "    description: Manage the platform\n"
+        f"{cli_block}"
         "  resources:\n"
         "  - settings.yaml\n"
         "  componentFiles:\n" + "\n".join(comp_blocks) + "\n"
