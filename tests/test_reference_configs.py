@@ -107,3 +107,59 @@ def test_init_config_standalone_byte_identical(capsys):
     with open(os.path.join(CONFIGS, "standalone", "valid.yaml")) as f:
         golden = f.read()
     assert out == golden
+
+
+def test_init_config_collection_matches_reference_builder(capsys):
+    """init-config collection output pinned to the reference's sample
+    builder semantics (pkg/cli/init_config.go:133-146 + const names
+    :36-41): collectionSampleName, NewSampleAPISpec values, rootcmd
+    defaulted as non-sub, subcmd defaulted as sub (name `collection`)."""
+    assert main(["init-config", "collection"]) == 0
+    out = capsys.readouterr().out
+    assert out == (
+        "kind: WorkloadCollection\n"
+        "name: workload-collection-config\n"
+        "spec:\n"
+        "  api:\n"
+        "    clusterScoped: false\n"
+        "    domain: acme.com\n"
+        "    group: apps\n"
+        "    kind: MyApp\n"
+        "    version: v1alpha1\n"
+        "  companionCliRootcmd:\n"
+        "    description: Manage myapp collection and components\n"
+        "    name: myapp\n"
+        "  companionCliSubcmd:\n"
+        "    description: Manage myapp workload\n"
+        "    name: collection\n"
+        "  componentFiles:\n"
+        "    - /path/to/my/component-workload-config.yaml\n"
+        "  resources:\n"
+        "    - /path/to/my/child-resources.yaml\n"
+    )
+
+
+def test_init_config_component_matches_reference_builder(capsys):
+    """init-config component output pinned to the reference's sample
+    builder (pkg/cli/init_config.go:147-158): componentSampleName,
+    dependency on `<name>-2`, subcmd defaulted as sub."""
+    assert main(["init-config", "component"]) == 0
+    out = capsys.readouterr().out
+    assert out == (
+        "kind: ComponentWorkload\n"
+        "name: component-workload-config\n"
+        "spec:\n"
+        "  api:\n"
+        "    clusterScoped: false\n"
+        "    domain: acme.com\n"
+        "    group: apps\n"
+        "    kind: MyApp\n"
+        "    version: v1alpha1\n"
+        "  companionCliSubcmd:\n"
+        "    description: Manage myapp workload\n"
+        "    name: myapp\n"
+        "  dependencies:\n"
+        "    - component-workload-config-2\n"
+        "  resources:\n"
+        "    - /path/to/my/component-workload-config.yaml\n"
+    )
