@@ -114,6 +114,8 @@ def test_init_config_collection_matches_reference_builder(capsys):
     builder semantics (pkg/cli/init_config.go:133-146 + const names
     :36-41): collectionSampleName, NewSampleAPISpec values, rootcmd
     defaulted as non-sub, subcmd defaulted as sub (name `collection`)."""
+    from operator_builder_amd.cli.main import main
+
     assert main(["init-config", "collection"]) == 0
     out = capsys.readouterr().out
     assert out == (
@@ -143,6 +145,8 @@ def test_init_config_component_matches_reference_builder(capsys):
     """init-config component output pinned to the reference's sample
     builder (pkg/cli/init_config.go:147-158): componentSampleName,
     dependency on `<name>-2`, subcmd defaulted as sub."""
+    from operator_builder_amd.cli.main import main
+
     assert main(["init-config", "component"]) == 0
     out = capsys.readouterr().out
     assert out == (
