@@ -113,7 +113,18 @@ def run(args) -> None:
         backend = "nccl" if use_cuda else "gloo"
         if use_cuda:
             torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
-        dist.init_process_group(backend=backend)
+        # gloo's mesh-connect writes "[Gloo] Rank ..." straight to the C
+        # stdout; divert fd 1 to stderr around init so stdout carries
+        # ONLY the single JSON record the driver contract requires
+        sys.stdout.flush()
+        saved_stdout = os.dup(1)
+        try:
+            os.dup2(2, 1)
+            dist.init_process_group(backend=backend)
+        finally:
+            sys.stdout.flush()
+            os.dup2(saved_stdout, 1)
+            os.close(saved_stdout)
 
     def barrier_sync():
         if distributed:
