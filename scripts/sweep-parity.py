@@ -51,6 +51,8 @@ def main() -> int:
          m.test_edge_semantics_stay_byte_identical),
         ("upgrade-sequences", m.upgrade_plans,
          m.test_random_upgrade_sequences_stay_byte_identical),
+        ("nested-structures", m.structure_setups,
+         m.test_random_structures_stay_byte_identical),
     ]
     for name, strategy, test in suites:
         inner = test.hypothesis.inner_test

@@ -583,3 +583,143 @@ def test_random_upgrade_sequences_stay_byte_identical(
         if r.missing or r.diff_lines != 0
     }
     assert not bad, f"diverged for plan {plan}:\n{bad}"
+
+
+def _yaml_value(draw, depth):
+    kind = draw(
+        st.sampled_from(
+            ["str", "int", "bool", "map", "list"]
+            if depth < 3
+            else ["str", "int", "bool"]
+        )
+    )
+    if kind == "str":
+        return f'"{draw(names)}"'
+    if kind == "int":
+        return str(draw(st.integers(min_value=0, max_value=9999)))
+    if kind == "bool":
+        return "true" if draw(st.booleans()) else "false"
+    if kind == "map":
+        n = draw(st.integers(min_value=1, max_value=3))
+        keys = draw(
+            st.lists(names, min_size=n, max_size=n, unique=True)
+        )
+        return {k: _yaml_value(draw, depth + 1) for k in keys}
+    n = draw(st.integers(min_value=1, max_value=3))
+    return [_yaml_value(draw, depth + 1) for _ in range(n)]
+
+
+def _render_yaml(value, indent):
+    pad = "  " * indent
+    if isinstance(value, dict):
+        lines = []
+        for k, v in value.items():
+            if isinstance(v, (dict, list)):
+                lines.append(f"{pad}{k}:")
+                lines.append(_render_yaml(v, indent + 1))
+            else:
+                lines.append(f"{pad}{k}: {v}")
+        return "\n".join(lines)
+    if isinstance(value, list):
+        lines = []
+        for v in value:
+            if isinstance(v, dict):
+                inner = _render_yaml(v, indent + 1).split("\n")
+                first = inner[0].strip()
+                lines.append(f"{pad}- {first}")
+                lines.extend(inner[1:])
+            elif isinstance(v, list):
+                lines.append(f"{pad}-")
+                lines.append(_render_yaml(v, indent + 1))
+            else:
+                lines.append(f"{pad}- {v}")
+        return "\n".join(lines)
+    return f"{pad}{value}"
+
+
+@st.composite
+def structure_setups(draw):
+    """Random nested YAML manifest bodies — stress the YAML round-trip,
+    the object code generator, and the Go re-indenter together."""
+    group = draw(kinds_names)
+    kind = draw(kinds_names).capitalize() + "Deep"
+    field = draw(names)
+    tree = _yaml_value(draw, 0)
+    while not isinstance(tree, dict):
+        tree = _yaml_value(draw, 0)
+
+    config = (
+        f"name: {draw(names)}-deep\n"
+        "kind: StandaloneWorkload\n"
+        "spec:\n"
+        "  api:\n"
+        "    domain: example.com\n"
+        f"    group: {group}\n"
+        "    version: v1alpha1\n"
+        f"    kind: {kind}\n"
+        "    clusterScoped: false\n"
+        "  resources:\n"
+        "  - r.yaml\n"
+    )
+    manifest = (
+        "apiVersion: v1\n"
+        "kind: ConfigMap\n"
+        "metadata:\n"
+        "  name: deep-config\n"
+        "  namespace: default\n"
+        "data:\n"
+        f"  # +operator-builder:field:name={field},type=string,"
+        'default="x"\n'
+        '  marked: "x"\n'
+        "extra:\n" + _render_yaml(tree, 1) + "\n"
+    )
+    return config, manifest
+
+
+@settings(
+    max_examples=10,
+    deadline=None,
+    derandomize=True,
+    suppress_health_check=list(HealthCheck),
+)
+@given(structure_setups())
+def test_random_structures_stay_byte_identical(tmp_path_factory, setup):
+    config, manifest = setup
+    workdir = tmp_path_factory.mktemp("parityfuzzdeep")
+    cfg_dir = workdir / ".workloadConfig"
+    cfg_dir.mkdir()
+    (cfg_dir / "workload.yaml").write_text(config)
+    (cfg_dir / "r.yaml").write_text(manifest)
+
+    cwd = os.getcwd()
+    os.chdir(workdir)
+    try:
+        assert (
+            main(
+                [
+                    "init",
+                    "--workload-config",
+                    ".workloadConfig/workload.yaml",
+                    "--repo",
+                    "github.com/fuzz/deep",
+                ]
+            )
+            == 0
+        ), manifest
+        assert main(["create", "api"]) == 0, manifest
+
+        project = Project.load(".")
+        processor = workload_config.parse(".workloadConfig/workload.yaml")
+        subcommand.create_api(processor)
+        ctx = _build_context(".", project, processor.workload)
+        report = diff_report(".", ctx, processor.workload)
+    finally:
+        os.chdir(cwd)
+        shutil.rmtree(workdir, ignore_errors=True)
+
+    bad = {
+        r.path: ("MISSING" if r.missing else r.diff_lines)
+        for r in report
+        if r.missing or r.diff_lines != 0
+    }
+    assert not bad, f"diverged for:\n{manifest}\n{bad}"
