@@ -77,6 +77,12 @@ def workload_setups(draw):
             "    description: Manage the workload\n"
         )
 
+    # one or two manifest files (multi-file exercises per-file
+    # definition outputs and the resources list plumbing)
+    two_files = draw(st.booleans())
+    resources_block = "  resources:\n  - r.yaml\n"
+    if two_files:
+        resources_block += "  - r2.yaml\n"
     config = (
         f"name: {draw(kinds_names)}-workload\n"
         "kind: StandaloneWorkload\n"
@@ -88,8 +94,7 @@ def workload_setups(draw):
         f"    kind: {kind}\n"
         f"    clusterScoped: {'true' if cluster_scoped else 'false'}\n"
         f"{cli_block}"
-        "  resources:\n"
-        "  - r.yaml\n"
+        f"{resources_block}"
     )
 
     if resource_kind == "ConfigMap":
@@ -139,7 +144,19 @@ def workload_setups(draw):
             "    - port: 80\n"
         )
 
-    return config, manifest
+    extra = None
+    if two_files:
+        extra = (
+            "apiVersion: v1\n"
+            "kind: Secret\n"
+            "metadata:\n"
+            f"  name: {draw(names)}-secret\n"
+            "  namespace: default\n"
+            "type: Opaque\n"
+            "stringData:\n"
+            f'  token: "{draw(names)}"\n'
+        )
+    return config, manifest, extra
 
 
 @settings(
@@ -150,12 +167,14 @@ def workload_setups(draw):
 )
 @given(workload_setups())
 def test_random_workloads_stay_byte_identical(tmp_path_factory, setup):
-    config, manifest = setup
+    config, manifest, extra = setup
     workdir = tmp_path_factory.mktemp("parityfuzz")
     cfg_dir = workdir / ".workloadConfig"
     cfg_dir.mkdir()
     (cfg_dir / "workload.yaml").write_text(config)
     (cfg_dir / "r.yaml").write_text(manifest)
+    if extra is not None:
+        (cfg_dir / "r2.yaml").write_text(extra)
 
     cwd = os.getcwd()
     os.chdir(workdir)
